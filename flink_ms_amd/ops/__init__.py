@@ -1,0 +1,195 @@
+"""Op dispatch: HIP kernels on GPU, torch reference on CPU.
+
+On a GPU box the HIP extension (``flink_ms_amd/_hip_ops.so``, built in-tree
+by ``flink_ms_amd.ops.build``) is REQUIRED: a missing extension raises
+instead of silently falling back to eager torch, so benchmarks and GPU tests
+always exercise the native path.  On CPU-only machines the pure-torch
+reference implementations (``flink_ms_amd.ops.reference``) run instead.
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+from ..data.blocked import CSR
+from . import reference
+
+_hip_ops = None
+_hip_import_error: Optional[BaseException] = None
+try:
+    from flink_ms_amd import _hip_ops  # type: ignore[no-redef]
+except BaseException as e:  # pragma: no cover - exercised on unbuilt trees
+    _hip_import_error = e
+
+
+def hip_available() -> bool:
+    return _hip_ops is not None
+
+
+def _require_hip():
+    if _hip_ops is None:
+        raise RuntimeError(
+            "flink_ms_amd HIP extension is not built but a GPU tensor was "
+            "passed; run `python -m flink_ms_amd.ops.build` "
+            f"(import error: {_hip_import_error!r})"
+        )
+    return _hip_ops
+
+
+def _stream() -> int:
+    return torch.cuda.current_stream().cuda_stream
+
+
+def _pad_k(t: torch.Tensor) -> torch.Tensor:
+    """Pad factor columns to the next multiple of 16 (kernel tile width)."""
+    k = t.shape[1]
+    kp = ((k + 15) // 16) * 16
+    if kp == k:
+        return t
+    out = torch.zeros(t.shape[0], kp, dtype=t.dtype, device=t.device)
+    out[:, :k] = t
+    return out
+
+
+_EMPTY = {}
+
+
+def _empty(device) -> torch.Tensor:
+    key = str(device)
+    if key not in _EMPTY:
+        _EMPTY[key] = torch.empty(0, device=device)
+    return _EMPTY[key]
+
+
+# -------------------------------------------------------------------- ALS
+
+def als_solve_side(
+    csr: CSR,
+    other_factors: torch.Tensor,
+    reg: float,
+    out_bf16: Optional[torch.Tensor] = None,
+    row_order: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    """One ALS half-iteration: solve every row entity of ``csr`` against the
+    opposite side's factors.  Returns fp32 [num_rows, k]; optionally also
+    writes the bf16 image for the next half-iteration in the same kernel."""
+    if other_factors.is_cuda:
+        ops = _require_hip()
+        fac = _pad_k(other_factors.to(torch.bfloat16).contiguous())
+        k = fac.shape[1]
+        out = torch.empty(csr.num_rows, k, dtype=torch.float32,
+                          device=fac.device)
+        ob = out_bf16 if out_bf16 is not None else _empty(fac.device)
+        ro = row_order if row_order is not None else _empty(fac.device)
+        ops.als_solve_fused(csr.indptr, csr.indices, csr.values, fac,
+                            out, ob, ro, float(reg), _stream())
+        korig = other_factors.shape[1]
+        return out[:, :korig] if korig != k else out
+    return reference.als_solve_side_reference(csr, other_factors, reg)
+
+
+def gramian(csr: CSR, factors: torch.Tensor, reg: float) -> Tuple[torch.Tensor, torch.Tensor]:
+    if factors.is_cuda:
+        ops = _require_hip()
+        fac = _pad_k(factors.to(torch.bfloat16).contiguous())
+        k = fac.shape[1]
+        A = torch.empty(csr.num_rows, k, k, dtype=torch.float32, device=fac.device)
+        b = torch.empty(csr.num_rows, k, dtype=torch.float32, device=fac.device)
+        ops.gramian(csr.indptr, csr.indices, csr.values, fac, A, b,
+                    float(reg), _stream())
+        korig = factors.shape[1]
+        if korig != k:
+            return A[:, :korig, :korig], b[:, :korig]
+        return A, b
+    return reference.gramian_reference(csr, factors, reg)
+
+
+def cholesky_solve(A: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    if A.is_cuda:
+        ops = _require_hip()
+        A = A.contiguous()
+        b = b.contiguous()
+        x = torch.empty_like(b)
+        ops.cholesky_solve(A, b, x, _stream())
+        return x
+    return reference.cholesky_solve_reference(A, b)
+
+
+# -------------------------------------------------------------------- SVM
+
+def sdca_pass(
+    csr: CSR,
+    y: torch.Tensor,
+    alpha: torch.Tensor,
+    v: torch.Tensor,
+    lamb: float,
+    n_global: int,
+    norms_sq: Optional[torch.Tensor] = None,
+    perm: Optional[torch.Tensor] = None,
+) -> None:
+    """One local SDCA pass over the shard, updating alpha and v in place."""
+    if v.is_cuda:
+        ops = _require_hip()
+        if norms_sq is None:
+            norms_sq = csr_row_norms_sq(csr)
+        p = perm.to(torch.int32) if perm is not None else _empty(v.device)
+        ops.sdca_pass(csr.indptr, csr.indices, csr.values, y, norms_sq, p,
+                      alpha, v, 1.0 / (lamb * n_global), _stream())
+        return
+    reference.sdca_epoch_reference(csr, y, alpha, v, lamb, n_global, perm)
+
+
+def csr_row_norms_sq(csr: CSR) -> torch.Tensor:
+    row_ids = torch.repeat_interleave(
+        torch.arange(csr.num_rows, dtype=torch.int64, device=csr.device),
+        csr.row_counts(),
+    )
+    out = torch.zeros(csr.num_rows, dtype=torch.float32, device=csr.device)
+    out.index_add_(0, row_ids, csr.values * csr.values)
+    return out
+
+
+def svm_margins(csr: CSR, w: torch.Tensor) -> torch.Tensor:
+    if w.is_cuda:
+        ops = _require_hip()
+        out = torch.empty(csr.num_rows, dtype=torch.float32, device=w.device)
+        ops.svm_margins(csr.indptr, csr.indices, csr.values, w.contiguous(),
+                        out, _stream())
+        return out
+    return reference.svm_margins_reference(csr, w)
+
+
+# ---------------------------------------------------------------- serving
+
+def predict_dot(U: torch.Tensor, V: torch.Tensor, u_idx: torch.Tensor,
+                i_idx: torch.Tensor) -> torch.Tensor:
+    if U.is_cuda:
+        ops = _require_hip()
+        out = torch.empty(u_idx.numel(), dtype=torch.float32, device=U.device)
+        ops.predict_dot(U.contiguous(), V.contiguous(), u_idx.long(),
+                        i_idx.long(), out, _stream())
+        return out
+    return reference.predict_dot_reference(U, V, u_idx, i_idx)
+
+
+def sgd_update(U: torch.Tensor, V: torch.Tensor, u_idx: torch.Tensor,
+               i_idx: torch.Tensor, r: torch.Tensor, lr: float,
+               user_reg: float, item_reg: float) -> torch.Tensor:
+    """In-place online SGD step on the bf16 factor store; returns errors."""
+    if U.is_cuda:
+        ops = _require_hip()
+        err = torch.empty(u_idx.numel(), dtype=torch.float32, device=U.device)
+        ops.sgd_update(U, V, u_idx.long(), i_idx.long(), r, err,
+                       float(lr), float(user_reg), float(item_reg), _stream())
+        return err
+    u = u_idx.long()
+    i = i_idx.long()
+    p = U[u].to(torch.float32)
+    q = V[i].to(torch.float32)
+    p_new, q_new, err = reference.sgd_update_reference(
+        p, q, r, lr, user_reg, item_reg)
+    U[u] = p_new.to(U.dtype)
+    V[i] = q_new.to(V.dtype)
+    return err

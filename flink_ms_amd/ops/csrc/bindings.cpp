@@ -1,0 +1,238 @@
+// Python bindings for the flink_ms_amd HIP kernels.
+//
+// This translation unit is deliberately HIP-header-free: kernels live in the
+// .hip files behind extern "C" launchers (enum hipError_t <-> int,
+// hipStream_t <-> void*), so only hipcc ever sees device code and this file
+// only sees torch tensors.  The current HIP stream is passed in from Python
+// (torch.cuda.current_stream().cuda_stream).
+
+#include <torch/extension.h>
+
+#include <cstdint>
+
+extern "C" {
+int fma_als_solve_fused(int k, const int64_t* indptr, const int* indices,
+                        const float* values, const unsigned short* factors,
+                        float* out_f32, unsigned short* out_bf16,
+                        const int* row_order, int64_t nrows, float reg,
+                        void* stream);
+int fma_gramian(int k, const int64_t* indptr, const int* indices,
+                const float* values, const unsigned short* factors,
+                float* A_out, float* b_out, int64_t nrows, float reg,
+                void* stream);
+int fma_cholesky_solve(int k, const float* A_in, const float* b_in,
+                       float* x_out, int64_t nrows, void* stream);
+int fma_sdca_pass(const int64_t* indptr, const int* indices,
+                  const float* values, const float* y, const float* norms_sq,
+                  const int* perm, float* alpha, float* v, int64_t nrows,
+                  float scale, void* stream);
+int fma_svm_margins(const int64_t* indptr, const int* indices,
+                    const float* values, const float* w, float* out,
+                    int64_t nrows, void* stream);
+int fma_predict_dot(const unsigned short* U, const unsigned short* V,
+                    const int64_t* u_idx, const int64_t* i_idx,
+                    float* out, int64_t nq, int k, void* stream);
+int fma_sgd_update(unsigned short* U, unsigned short* V,
+                   const int64_t* u_idx, const int64_t* i_idx,
+                   const float* r, float* err_out, int64_t nq, int k,
+                   float lr, float user_reg, float item_reg, void* stream);
+int fma_mfma_probe_f32(const float* A, const float* B, float* D, void* stream);
+int fma_mfma_probe_bf16(const unsigned short* Xt, const unsigned short* Yt,
+                        float* C, void* stream);
+const char* fma_err_str(int err);
+}
+
+namespace {
+
+void check_hip(int err, const char* what) {
+    TORCH_CHECK(err == 0, what, " failed: ", fma_err_str(err));
+}
+
+void check_t(const torch::Tensor& t, torch::ScalarType dt, const char* name) {
+    TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+    TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+    TORCH_CHECK(t.scalar_type() == dt, name, " has wrong dtype");
+}
+
+const unsigned short* bf16_ptr(const torch::Tensor& t) {
+    return reinterpret_cast<const unsigned short*>(t.data_ptr());
+}
+unsigned short* bf16_ptr_mut(torch::Tensor& t) {
+    return reinterpret_cast<unsigned short*>(t.data_ptr());
+}
+
+void als_solve_fused(torch::Tensor indptr, torch::Tensor indices,
+                     torch::Tensor values, torch::Tensor factors,
+                     torch::Tensor out_f32, torch::Tensor out_bf16,
+                     torch::Tensor row_order, double reg, int64_t stream) {
+    check_t(indptr, torch::kInt64, "indptr");
+    check_t(indices, torch::kInt32, "indices");
+    check_t(values, torch::kFloat32, "values");
+    check_t(factors, torch::kBFloat16, "factors");
+    check_t(out_f32, torch::kFloat32, "out_f32");
+    const int k = (int)factors.size(1);
+    const long long nrows = indptr.size(0) - 1;
+    TORCH_CHECK(out_f32.size(0) == nrows && out_f32.size(1) == k,
+                "out_f32 shape mismatch");
+    unsigned short* ob = nullptr;
+    if (out_bf16.numel() > 0) {
+        check_t(out_bf16, torch::kBFloat16, "out_bf16");
+        TORCH_CHECK(out_bf16.numel() == out_f32.numel(), "out_bf16 shape");
+        ob = bf16_ptr_mut(out_bf16);
+    }
+    const int* order = nullptr;
+    if (row_order.numel() > 0) {
+        check_t(row_order, torch::kInt32, "row_order");
+        TORCH_CHECK(row_order.numel() == nrows, "row_order size");
+        order = row_order.data_ptr<int>();
+    }
+    check_hip(fma_als_solve_fused(
+                  k, indptr.data_ptr<int64_t>(), indices.data_ptr<int>(),
+                  values.data_ptr<float>(), bf16_ptr(factors),
+                  out_f32.data_ptr<float>(), ob, order, nrows, (float)reg,
+                  (void*)stream),
+              "als_solve_fused");
+}
+
+void gramian(torch::Tensor indptr, torch::Tensor indices, torch::Tensor values,
+             torch::Tensor factors, torch::Tensor A_out, torch::Tensor b_out,
+             double reg, int64_t stream) {
+    check_t(indptr, torch::kInt64, "indptr");
+    check_t(indices, torch::kInt32, "indices");
+    check_t(values, torch::kFloat32, "values");
+    check_t(factors, torch::kBFloat16, "factors");
+    check_t(A_out, torch::kFloat32, "A_out");
+    check_t(b_out, torch::kFloat32, "b_out");
+    const int k = (int)factors.size(1);
+    const long long nrows = indptr.size(0) - 1;
+    check_hip(fma_gramian(k, indptr.data_ptr<int64_t>(),
+                          indices.data_ptr<int>(), values.data_ptr<float>(),
+                          bf16_ptr(factors), A_out.data_ptr<float>(),
+                          b_out.data_ptr<float>(), nrows, (float)reg,
+                          (void*)stream),
+              "gramian");
+}
+
+void cholesky_solve(torch::Tensor A, torch::Tensor b, torch::Tensor x,
+                    int64_t stream) {
+    check_t(A, torch::kFloat32, "A");
+    check_t(b, torch::kFloat32, "b");
+    check_t(x, torch::kFloat32, "x");
+    const long long nrows = A.size(0);
+    const int k = (int)A.size(1);
+    check_hip(fma_cholesky_solve(k, A.data_ptr<float>(), b.data_ptr<float>(),
+                                 x.data_ptr<float>(), nrows, (void*)stream),
+              "cholesky_solve");
+}
+
+void sdca_pass(torch::Tensor indptr, torch::Tensor indices,
+               torch::Tensor values, torch::Tensor y, torch::Tensor norms_sq,
+               torch::Tensor perm, torch::Tensor alpha, torch::Tensor v,
+               double scale, int64_t stream) {
+    check_t(indptr, torch::kInt64, "indptr");
+    check_t(indices, torch::kInt32, "indices");
+    check_t(values, torch::kFloat32, "values");
+    check_t(y, torch::kFloat32, "y");
+    check_t(norms_sq, torch::kFloat32, "norms_sq");
+    check_t(alpha, torch::kFloat32, "alpha");
+    check_t(v, torch::kFloat32, "v");
+    const int* p = nullptr;
+    if (perm.numel() > 0) {
+        check_t(perm, torch::kInt32, "perm");
+        p = perm.data_ptr<int>();
+    }
+    check_hip(fma_sdca_pass(indptr.data_ptr<int64_t>(),
+                            indices.data_ptr<int>(), values.data_ptr<float>(),
+                            y.data_ptr<float>(), norms_sq.data_ptr<float>(),
+                            p, alpha.data_ptr<float>(), v.data_ptr<float>(),
+                            indptr.size(0) - 1, (float)scale, (void*)stream),
+              "sdca_pass");
+}
+
+void svm_margins(torch::Tensor indptr, torch::Tensor indices,
+                 torch::Tensor values, torch::Tensor w, torch::Tensor out,
+                 int64_t stream) {
+    check_t(indptr, torch::kInt64, "indptr");
+    check_t(indices, torch::kInt32, "indices");
+    check_t(values, torch::kFloat32, "values");
+    check_t(w, torch::kFloat32, "w");
+    check_t(out, torch::kFloat32, "out");
+    check_hip(fma_svm_margins(indptr.data_ptr<int64_t>(),
+                              indices.data_ptr<int>(),
+                              values.data_ptr<float>(), w.data_ptr<float>(),
+                              out.data_ptr<float>(), indptr.size(0) - 1,
+                              (void*)stream),
+              "svm_margins");
+}
+
+void predict_dot(torch::Tensor U, torch::Tensor V, torch::Tensor u_idx,
+                 torch::Tensor i_idx, torch::Tensor out, int64_t stream) {
+    check_t(U, torch::kBFloat16, "U");
+    check_t(V, torch::kBFloat16, "V");
+    check_t(u_idx, torch::kInt64, "u_idx");
+    check_t(i_idx, torch::kInt64, "i_idx");
+    check_t(out, torch::kFloat32, "out");
+    TORCH_CHECK(U.size(1) == V.size(1), "rank mismatch");
+    check_hip(fma_predict_dot(bf16_ptr(U), bf16_ptr(V),
+                              u_idx.data_ptr<int64_t>(),
+                              i_idx.data_ptr<int64_t>(),
+                              out.data_ptr<float>(), u_idx.numel(),
+                              (int)U.size(1), (void*)stream),
+              "predict_dot");
+}
+
+void sgd_update(torch::Tensor U, torch::Tensor V, torch::Tensor u_idx,
+                torch::Tensor i_idx, torch::Tensor r, torch::Tensor err_out,
+                double lr, double user_reg, double item_reg, int64_t stream) {
+    check_t(U, torch::kBFloat16, "U");
+    check_t(V, torch::kBFloat16, "V");
+    check_t(u_idx, torch::kInt64, "u_idx");
+    check_t(i_idx, torch::kInt64, "i_idx");
+    check_t(r, torch::kFloat32, "r");
+    float* ep = nullptr;
+    if (err_out.numel() > 0) {
+        check_t(err_out, torch::kFloat32, "err_out");
+        ep = err_out.data_ptr<float>();
+    }
+    check_hip(fma_sgd_update(bf16_ptr_mut(U), bf16_ptr_mut(V),
+                             u_idx.data_ptr<int64_t>(),
+                             i_idx.data_ptr<int64_t>(), r.data_ptr<float>(),
+                             ep, u_idx.numel(), (int)U.size(1), (float)lr,
+                             (float)user_reg, (float)item_reg, (void*)stream),
+              "sgd_update");
+}
+
+void mfma_probe_f32(torch::Tensor A, torch::Tensor B, torch::Tensor D,
+                    int64_t stream) {
+    check_t(A, torch::kFloat32, "A");
+    check_t(B, torch::kFloat32, "B");
+    check_t(D, torch::kFloat32, "D");
+    check_hip(fma_mfma_probe_f32(A.data_ptr<float>(), B.data_ptr<float>(),
+                                 D.data_ptr<float>(), (void*)stream),
+              "mfma_probe_f32");
+}
+
+void mfma_probe_bf16(torch::Tensor Xt, torch::Tensor Yt, torch::Tensor C,
+                     int64_t stream) {
+    check_t(Xt, torch::kBFloat16, "Xt");
+    check_t(Yt, torch::kBFloat16, "Yt");
+    check_t(C, torch::kFloat32, "C");
+    check_hip(fma_mfma_probe_bf16(bf16_ptr(Xt), bf16_ptr(Yt),
+                                  C.data_ptr<float>(), (void*)stream),
+              "mfma_probe_bf16");
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.doc() = "flink_ms_amd MI355X (gfx950) HIP kernels";
+    m.def("als_solve_fused", &als_solve_fused);
+    m.def("gramian", &gramian);
+    m.def("cholesky_solve", &cholesky_solve);
+    m.def("sdca_pass", &sdca_pass);
+    m.def("svm_margins", &svm_margins);
+    m.def("predict_dot", &predict_dot);
+    m.def("sgd_update", &sgd_update);
+    m.def("mfma_probe_f32", &mfma_probe_f32);
+    m.def("mfma_probe_bf16", &mfma_probe_bf16);
+}

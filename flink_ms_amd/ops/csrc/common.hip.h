@@ -1,0 +1,48 @@
+// Common device helpers for the flink_ms_amd CDNA4 (gfx950) kernels.
+// Single HIP path, MI355X-only: wave64, MFMA bf16 16x16x32 tiles, LDS staging.
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#define WAVE 64
+#define DEV_INLINE __device__ __forceinline__
+
+// MFMA fragment/accumulator register types (cdna_hip_programming.md §3):
+// 8 bf16 (4 VGPRs) per A/B fragment, 4 fp32 accumulators per 16x16 C/D tile.
+typedef short bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+DEV_INLINE float bf2f(unsigned short u) {
+    union { unsigned int i; float f; } v;
+    v.i = ((unsigned int)u) << 16;
+    return v.f;
+}
+
+DEV_INLINE unsigned short f2bf(float f) {
+    union { float f; unsigned int i; } v;
+    v.f = f;
+    // round-to-nearest-even (matches torch bfloat16 casts for finite values)
+    unsigned int r = v.i + 0x7fffu + ((v.i >> 16) & 1u);
+    return (unsigned short)(r >> 16);
+}
+
+DEV_INLINE float wave_reduce_sum(float x) {
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        x += __shfl_xor(x, off, WAVE);
+    return x;
+}
+
+// Upper-triangle 16x16-tile enumeration for the Gramian accumulator:
+// tiles 0..KT*(KT+1)/2-1 cover (p,q) with p<=q row by row; the KT tiles after
+// that are the (p, EXT) rating-column tiles that produce b.
+constexpr int up_tile_p(int t, int KT) {
+    int p = 0;
+    while (t >= KT - p) { t -= KT - p; ++p; }
+    return p;
+}
+constexpr int up_tile_q(int t, int KT) {
+    int p = 0;
+    while (t >= KT - p) { t -= KT - p; ++p; }
+    return p + t;
+}
