@@ -1,0 +1,172 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: ALS training throughput on MI355X (BASELINE.json).
+
+Metric: ALS ratings/sec per iteration on synthetic MovieLens-25M-shape
+ratings at rank 64 bf16 (BASELINE.json config 2; config 4 via --rank 128
+--scale-1b).  Weak scaling: every rank owns its own 25M-shape user shard
+(162,541 users x 25,000,095 ratings) against a shared 59,047-item catalog;
+item entities are range-sharded and each half-iteration all-gathers the
+opposite side's bf16 factors over xGMI (RCCL).
+
+Driver contract:
+  python bench.py --gpus N --steps K --warmup W
+  (N>1 is launched by the driver via torch.distributed.run; this script reads
+  RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* from the env.)
+Rank 0 prints ONE JSON line; timing is W untimed warmups then EXACTLY K
+steps bracketed by barrier + torch.cuda.synchronize on both sides, MAX over
+ranks.  --svm additionally reports the CoCoA-SVM samples/sec secondary
+metric in config.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import sys
+import time
+
+import torch
+
+from flink_ms_amd.data.libsvm import LibSVMShape, synthetic_libsvm
+from flink_ms_amd.data.ratings import ML25M_SHAPE, RatingsShape, synthetic_ratings
+from flink_ms_amd.models.als import ALSConfig, ALSTrainer
+from flink_ms_amd.models.svm import SVMConfig, SVMTrainer
+from flink_ms_amd.parallel.dist import init_from_env
+
+
+def parse_args(argv=None):
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1,
+                   help="expected world size (informational; actual from env)")
+    p.add_argument("--steps", type=int, default=8)
+    p.add_argument("--warmup", type=int, default=2)
+    p.add_argument("--rank", type=int, default=64, help="latent factors")
+    p.add_argument("--ratings-per-gpu", type=int, default=ML25M_SHAPE.num_ratings)
+    p.add_argument("--users-per-gpu", type=int, default=ML25M_SHAPE.num_users)
+    p.add_argument("--items", type=int, default=ML25M_SHAPE.num_items)
+    p.add_argument("--lambda", dest="lambda_", type=float, default=0.9)
+    p.add_argument("--seed", type=int, default=42)
+    p.add_argument("--scale-1b", action="store_true",
+                   help="1B-rating config: 125M ratings x 1.25M users per GPU")
+    p.add_argument("--svm", action="store_true",
+                   help="also run the CoCoA-SVM secondary bench")
+    p.add_argument("--svm-rows-per-gpu", type=int, default=697_641)
+    p.add_argument("--device", default=None, help="cpu override for tests")
+    return p.parse_args(argv)
+
+
+def bench_als(args, ctx):
+    if args.scale_1b:
+        users_pg, ratings_pg, items = 1_250_000, 125_000_000, 500_000
+        model_name = "ALS-1B-shape"
+    else:
+        users_pg, ratings_pg, items = (args.users_per_gpu,
+                                       args.ratings_per_gpu, args.items)
+        model_name = "ALS-ml25m-shape"
+    world = ctx.world_size
+    num_users = users_pg * world
+
+    shape = RatingsShape(users_pg, items, ratings_pg)
+    u, i, r = synthetic_ratings(shape, seed=args.seed + ctx.rank)
+    u = (u.long() + ctx.rank * users_pg).to(torch.int64)  # global user ids
+
+    cfg = ALSConfig(iterations=args.steps, num_factors=args.rank,
+                    lambda_=args.lambda_, seed=args.seed,
+                    dtype=torch.bfloat16 if ctx.device.type == "cuda"
+                    else torch.float32)
+    trainer = ALSTrainer(cfg, ctx)
+    trainer.setup(u, i.long(), r, num_users, items)
+
+    for _ in range(args.warmup):
+        trainer.step()
+    ctx.barrier()
+    if ctx.device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        trainer.step()
+    ctx.barrier()
+    if ctx.device.type == "cuda":
+        torch.cuda.synchronize()
+    elapsed = ctx.max_scalar(time.perf_counter() - t0)
+
+    total_nnz = ratings_pg * world
+    ms_per_step = elapsed / args.steps * 1000.0
+    value = total_nnz / (elapsed / args.steps)
+    return {
+        "value": value,
+        "ms_per_step": ms_per_step,
+        "model": model_name,
+        "config": {
+            "model": model_name,
+            "rank": args.rank,
+            "ratings_per_gpu": ratings_pg,
+            "users_per_gpu": users_pg,
+            "num_items": items,
+            "global_batch": total_nnz,
+            "seq_len": None,
+            "parallelism": f"dp{world}+factor-allgather",
+        },
+    }
+
+
+def bench_svm(args, ctx):
+    shape = LibSVMShape(args.svm_rows_per_gpu, 47_236, 74)
+    csr, y = synthetic_libsvm(shape, seed=args.seed + ctx.rank)
+    cfg = SVMConfig(iterations=args.steps, local_iterations=1,
+                    regularization=0.01, seed=args.seed)
+    tr = SVMTrainer(cfg, ctx)
+    tr.setup(csr, y)
+    for _ in range(args.warmup):
+        tr.step()
+    ctx.barrier()
+    if ctx.device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        tr.step()
+    ctx.barrier()
+    if ctx.device.type == "cuda":
+        torch.cuda.synchronize()
+    elapsed = ctx.max_scalar(time.perf_counter() - t0)
+    samples = shape.num_rows * ctx.world_size * cfg.local_iterations
+    return {"svm_samples_per_sec": samples / (elapsed / args.steps),
+            "svm_ms_per_step": elapsed / args.steps * 1000.0}
+
+
+def main(argv=None):
+    args = parse_args(argv)
+    if args.device == "cpu":
+        import os
+        os.environ.setdefault("WORLD_SIZE", "1")
+    ctx = init_from_env(backend="gloo" if args.device == "cpu" else None)
+    if args.device == "cpu":
+        ctx.device = torch.device("cpu")
+
+    als = bench_als(args, ctx)
+    extra = {}
+    if args.svm:
+        extra = bench_svm(args, ctx)
+
+    if ctx.rank == 0:
+        out = {
+            "metric": "ALS training throughput (ratings/sec per iteration)",
+            "value": als["value"],
+            "unit": "ratings/s",
+            "n_gpus": ctx.world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": als["ms_per_step"],
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "bf16" if ctx.device.type == "cuda" else "fp32",
+            "data": "synthetic",
+            "config": {**als["config"], **extra},
+        }
+        print(json.dumps(out), flush=True)
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

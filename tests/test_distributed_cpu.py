@@ -1,0 +1,141 @@
+"""Multi-process (gloo, world_size=2) tests of the distributed paths.
+
+These run on CPU-only machines and validate the collective structure the
+RCCL path shares: the one-time rating all-to-all (SURVEY.md C1 routing), the
+per-iteration factor all-gather, and CoCoA's delta-w all-reduce (C2) —
+distributed results must match the single-process reference.
+"""
+
+import multiprocessing as mp
+import os
+
+import pytest
+import torch
+
+WORLD = 2
+
+
+def _run_workers(target, extra=()):
+    ctx = mp.get_context("spawn")
+    port = 29500 + (os.getpid() % 500)
+    q = ctx.Queue()
+    procs = [ctx.Process(target=target, args=(rank, WORLD, port, q, *extra))
+             for rank in range(WORLD)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(WORLD):
+        rank, payload = q.get(timeout=180)
+        results[rank] = payload
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0, f"worker exit {p.exitcode}"
+    return results
+
+
+def _init(rank, world, port):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world), "LOCAL_RANK": str(rank),
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+    })
+    import flink_ms_amd.parallel.dist as D
+    D._CTX = None
+    return D.init_from_env(backend="gloo")
+
+
+def _als_worker(rank, world, port, q):
+    torch.manual_seed(0)
+    ctx = _init(rank, world, port)
+    from flink_ms_amd.data.ratings import RatingsShape, synthetic_ratings
+    from flink_ms_amd.models.als import ALSConfig, ALSTrainer
+    shape = RatingsShape(120, 60, 2000)
+    u, i, r = synthetic_ratings(shape, seed=5)
+    # split triples arbitrarily (by parity) — setup() re-exchanges by owner
+    mask = torch.arange(shape.num_ratings) % world == rank
+    cfg = ALSConfig(iterations=3, num_factors=8, lambda_=0.2,
+                    dtype=torch.float32)
+    tr = ALSTrainer(cfg, ctx)
+    tr.setup(u[mask].long(), i[mask].long(), r[mask],
+             shape.num_users, shape.num_items)
+    tr.fit()
+    m = tr.model()
+    q.put((rank, {
+        "user_ids": m.user_ids.tolist(),
+        "user_factors": m.user_factors.numpy().tolist(),
+        "item_ids": m.item_ids.tolist(),
+        "item_factors": m.item_factors.numpy().tolist(),
+    }))
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_distributed_als_matches_single_process():
+    results = _run_workers(_als_worker)
+    # single-process reference on the full triple set
+    from flink_ms_amd.data.ratings import RatingsShape, synthetic_ratings
+    from flink_ms_amd.models.als import ALSConfig, train_als
+    shape = RatingsShape(120, 60, 2000)
+    u, i, r = synthetic_ratings(shape, seed=5)
+    # match the sharded init: rank-dependent seeds per item shard
+    import flink_ms_amd.parallel.dist as D
+    D._CTX = None
+    model, _ = None, None
+    # Distributed init differs (per-rank seed), so compare MSE quality
+    # rather than exact factors.
+    from flink_ms_amd.models.mse import evaluate_mse
+    uf, itf, uids, iids = [], [], [], []
+    for rank in sorted(results):
+        uf += results[rank]["user_factors"]
+        itf += results[rank]["item_factors"]
+        uids += results[rank]["user_ids"]
+        iids += results[rank]["item_ids"]
+    U = torch.zeros(shape.num_users, 8)
+    V = torch.zeros(shape.num_items, 8)
+    U[torch.tensor(uids)] = torch.tensor(uf)
+    V[torch.tensor(iids)] = torch.tensor(itf)
+    res = evaluate_mse(U, V, u, i, r)
+    model_sp, _ = train_als(
+        u, i, r, shape.num_users, shape.num_items,
+        ALSConfig(iterations=3, num_factors=8, lambda_=0.2,
+                  dtype=torch.float32))
+    res_sp = evaluate_mse(model_sp.user_factors, model_sp.item_factors, u, i, r)
+    # distributed quality within 25% of single-process quality
+    assert res.mse < res_sp.mse * 1.25 + 0.05, (res.mse, res_sp.mse)
+
+
+def _svm_worker(rank, world, port, q):
+    ctx = _init(rank, world, port)
+    from flink_ms_amd.data.blocked import CSR
+    from flink_ms_amd.data.libsvm import LibSVMShape, synthetic_libsvm
+    from flink_ms_amd.models.svm import SVMConfig, SVMTrainer
+    csr, y = synthetic_libsvm(LibSVMShape(300, 40, 8), seed=9, separable=True)
+    # row-shard the one global dataset
+    lo, hi = rank * 150, (rank + 1) * 150
+    sub = CSR(csr.indptr[lo:hi + 1] - csr.indptr[lo],
+              csr.indices[csr.indptr[lo]:csr.indptr[hi]],
+              csr.values[csr.indptr[lo]:csr.indptr[hi]],
+              150, csr.num_cols)
+    tr = SVMTrainer(SVMConfig(iterations=4, local_iterations=2,
+                              regularization=0.01, seed=17 + rank), ctx)
+    tr.setup(sub, y[lo:hi])
+    assert tr.n_global == 300
+    model = tr.fit()
+    obj = tr.objective()
+    q.put((rank, {"w": model.weights.tolist(), "obj": obj}))
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_distributed_cocoa_svm():
+    results = _run_workers(_svm_worker)
+    w0 = torch.tensor(results[0]["w"])
+    w1 = torch.tensor(results[1]["w"])
+    # CoCoA's all-reduced w must be identical on every rank
+    assert torch.allclose(w0, w1, atol=1e-6)
+    # and must have learned something on the separable set
+    from flink_ms_amd.data.libsvm import LibSVMShape, synthetic_libsvm
+    from flink_ms_amd.ops import reference as R
+    csr, y = synthetic_libsvm(LibSVMShape(300, 40, 8), seed=9, separable=True)
+    margins = R.svm_margins_reference(csr, w0)
+    acc = float((margins.sign() == y).float().mean())
+    assert acc > 0.85, acc

@@ -1,0 +1,150 @@
+"""GPU parity tests: HIP kernels vs the plain-torch fp32 references."""
+
+import pytest
+import torch
+
+from flink_ms_amd import ops
+from flink_ms_amd.data.blocked import CSR, csr_from_coo
+from flink_ms_amd.ops import reference as R
+
+pytestmark = pytest.mark.gpu
+
+
+def _rand_csr(rows, cols, nnz, seed, device):
+    g = torch.Generator().manual_seed(seed)
+    r = torch.randint(0, rows, (nnz,), generator=g, dtype=torch.int32)
+    c = torch.randint(0, cols, (nnz,), generator=g, dtype=torch.int32)
+    v = torch.rand(nnz, generator=g) * 4.5 + 0.5
+    return csr_from_coo(r, c, v, rows, cols).to(device)
+
+
+@pytest.mark.parametrize("k", [16, 32, 64, 128])
+def test_gramian_parity(gpu, k):
+    csr = _rand_csr(rows=300, cols=200, nnz=20_000, seed=k, device=gpu)
+    fac_bf16 = (torch.randn(200, k, generator=torch.Generator().manual_seed(2))
+                * 0.5).to(torch.bfloat16)
+    A, b = ops.gramian(csr, fac_bf16.to(gpu), reg=0.7)
+    # reference accumulates the SAME bf16-quantized factors in fp32
+    A_ref, b_ref = R.gramian_reference(csr.to("cpu"), fac_bf16.to(torch.float32),
+                                       reg=0.7)
+    A_ref = A_ref.to(gpu)
+    b_ref = b_ref.to(gpu)
+    scale = A_ref.abs().amax()
+    assert (A - A_ref).abs().amax() / scale < 3e-3, \
+        f"A mismatch {(A - A_ref).abs().amax()} vs scale {scale}"
+    # b uses the bf16 hi/lo split of the rating: ~2^-16 relative error floor
+    bscale = b_ref.abs().amax()
+    assert (b - b_ref).abs().amax() / bscale < 3e-3
+
+
+@pytest.mark.parametrize("k", [16, 48, 64, 128])
+def test_cholesky_solve_parity(gpu, k):
+    g = torch.Generator().manual_seed(3)
+    B = 64
+    M = torch.randn(B, k, k, generator=g) * 0.3
+    A = (M @ M.transpose(1, 2) + 2.0 * torch.eye(k)).contiguous()
+    b = torch.randn(B, k, generator=g)
+    x = ops.cholesky_solve(A.to(gpu), b.to(gpu))
+    x_ref = R.cholesky_solve_reference(A, b).to(gpu)
+    torch.cuda.synchronize()
+    assert torch.allclose(x, x_ref, atol=1e-3, rtol=1e-3), \
+        (x - x_ref).abs().max()
+
+
+@pytest.mark.parametrize("k", [64, 128])
+def test_fused_solve_matches_modular(gpu, k):
+    csr = _rand_csr(rows=500, cols=300, nnz=40_000, seed=k + 1, device=gpu)
+    fac = (torch.randn(300, k, generator=torch.Generator().manual_seed(4))
+           * 0.5).to(torch.bfloat16).to(gpu)
+    out_fused = ops.als_solve_side(csr, fac, reg=0.9)
+    A, b = ops.gramian(csr, fac, reg=0.9)
+    out_mod = ops.cholesky_solve(A, b)
+    torch.cuda.synchronize()
+    assert torch.allclose(out_fused, out_mod, atol=1e-3, rtol=1e-3)
+    # and against the full fp32 reference
+    ref = R.als_solve_side_reference(csr.to("cpu"),
+                                     fac.cpu().to(torch.float32), reg=0.9)
+    err = (out_fused.cpu() - ref).abs().amax() / ref.abs().amax()
+    assert err < 2e-2, f"fused vs fp32 reference rel err {err}"
+
+
+def test_fused_solve_empty_rows(gpu):
+    # rows with no ratings must come back zero
+    indptr = torch.tensor([0, 0, 3, 3], dtype=torch.int64)
+    indices = torch.tensor([0, 1, 2], dtype=torch.int32)
+    values = torch.tensor([1.0, 2.0, 3.0])
+    csr = CSR(indptr, indices, values, 3, 4).to(gpu)
+    fac = torch.randn(4, 16).to(torch.bfloat16).to(gpu)
+    out = ops.als_solve_side(csr, fac, reg=0.5)
+    torch.cuda.synchronize()
+    assert torch.all(out[0] == 0) and torch.all(out[2] == 0)
+    assert torch.isfinite(out[1]).all() and out[1].abs().sum() > 0
+
+
+def test_sdca_kernel_converges_and_margins(gpu):
+    from flink_ms_amd.data.libsvm import LibSVMShape, synthetic_libsvm
+    csr, y = synthetic_libsvm(LibSVMShape(5000, 500, 20), seed=6,
+                              separable=True, device=str(gpu))
+    alpha = torch.zeros(5000, device=gpu)
+    v = torch.zeros(500, device=gpu)
+    lamb = 0.01
+    o0 = R.hinge_objective(csr.to("cpu"), y.cpu(), v.cpu(), lamb)
+    for _ in range(5):
+        ops.sdca_pass(csr, y, alpha, v, lamb, 5000)
+    torch.cuda.synchronize()
+    o1 = R.hinge_objective(csr.to("cpu"), y.cpu(), v.cpu(), lamb)
+    assert o1 < 0.7 * o0
+    assert (alpha >= 0).all() and (alpha <= 1).all()
+    # margins kernel parity
+    w = torch.randn(500, generator=torch.Generator().manual_seed(7)).to(gpu)
+    m = ops.svm_margins(csr, w)
+    m_ref = R.svm_margins_reference(csr.to("cpu"), w.cpu()).to(gpu)
+    assert torch.allclose(m, m_ref, atol=1e-3, rtol=1e-3)
+
+
+def test_predict_dot_parity(gpu):
+    g = torch.Generator().manual_seed(8)
+    U = (torch.randn(100, 64, generator=g)).to(torch.bfloat16).to(gpu)
+    V = (torch.randn(80, 64, generator=g)).to(torch.bfloat16).to(gpu)
+    u = torch.randint(0, 100, (500,), generator=g)
+    i = torch.randint(0, 80, (500,), generator=g)
+    out = ops.predict_dot(U, V, u.to(gpu), i.to(gpu))
+    ref = R.predict_dot_reference(U.cpu(), V.cpu(), u, i).to(gpu)
+    torch.cuda.synchronize()
+    assert torch.allclose(out, ref, atol=1e-2, rtol=1e-2)
+
+
+def test_sgd_update_parity(gpu):
+    g = torch.Generator().manual_seed(9)
+    U = (torch.randn(50, 64, generator=g) * 0.3).to(torch.bfloat16)
+    V = (torch.randn(40, 64, generator=g) * 0.3).to(torch.bfloat16)
+    # unique (u, i) pairs -> no in-batch collisions -> exact parity
+    u = torch.arange(0, 30, dtype=torch.int64)
+    i = torch.arange(0, 30, dtype=torch.int64)
+    r = torch.rand(30, generator=g) * 4.5 + 0.5
+    Ug, Vg = U.clone().to(gpu), V.clone().to(gpu)
+    err_g = ops.sgd_update(Ug, Vg, u.to(gpu), i.to(gpu), r.to(gpu),
+                           lr=0.1, user_reg=0.01, item_reg=0.02)
+    p, q = U[u].to(torch.float32), V[i].to(torch.float32)
+    pn, qn, err = R.sgd_update_reference(p, q, r, 0.1, 0.01, 0.02)
+    torch.cuda.synchronize()
+    assert torch.allclose(err_g.cpu(), err, atol=1e-2, rtol=1e-2)
+    assert torch.allclose(Ug[u.to(gpu)].cpu().to(torch.float32), pn,
+                          atol=2e-2, rtol=2e-2)
+    assert torch.allclose(Vg[i.to(gpu)].cpu().to(torch.float32), qn,
+                          atol=2e-2, rtol=2e-2)
+
+
+def test_als_trainer_gpu_converges(gpu):
+    from flink_ms_amd.data.ratings import RatingsShape, synthetic_ratings
+    from flink_ms_amd.models.als import ALSConfig, ALSTrainer
+    from flink_ms_amd.models.mse import evaluate_mse
+    shape = RatingsShape(3000, 1000, 100_000)
+    u, i, r = synthetic_ratings(shape, seed=11)
+    tr = ALSTrainer(ALSConfig(iterations=3, num_factors=64, lambda_=0.3))
+    tr.ctx.device = gpu
+    tr.setup(u.long(), i.long(), r, shape.num_users, shape.num_items)
+    tr.fit()
+    m = tr.model()
+    res = evaluate_mse(m.user_factors.to(gpu), m.item_factors.to(gpu), u, i, r)
+    assert res.mse < 1.2, f"GPU ALS MSE {res.mse}"

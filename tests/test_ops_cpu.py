@@ -1,0 +1,109 @@
+"""CPU tests of the reference ops + data layer (the HIP kernels' oracle)."""
+
+import torch
+
+from flink_ms_amd.data.blocked import csr_from_coo, csr_transpose
+from flink_ms_amd.ops import reference as R
+
+
+def _rand_csr(rows=50, cols=30, nnz=400, seed=0):
+    g = torch.Generator().manual_seed(seed)
+    r = torch.randint(0, rows, (nnz,), generator=g, dtype=torch.int32)
+    c = torch.randint(0, cols, (nnz,), generator=g, dtype=torch.int32)
+    v = torch.rand(nnz, generator=g) * 4.5 + 0.5
+    return csr_from_coo(r, c, v, rows, cols)
+
+
+def test_csr_roundtrip():
+    csr = _rand_csr()
+    assert csr.nnz == 400
+    t = csr_transpose(csr)
+    assert t.num_rows == 30 and t.num_cols == 50 and t.nnz == 400
+    tt = csr_transpose(t)
+    # transpose twice = same matrix (compare dense images)
+    def dense(c):
+        d = torch.zeros(c.num_rows, c.num_cols)
+        rows = torch.repeat_interleave(torch.arange(c.num_rows), c.row_counts())
+        d.index_put_((rows, c.indices.long()), c.values, accumulate=True)
+        return d
+    assert torch.allclose(dense(csr), dense(tt))
+
+
+def test_gramian_matches_dense():
+    csr = _rand_csr(rows=20, cols=15, nnz=150)
+    k = 8
+    fac = torch.randn(15, k)
+    A, b = R.gramian_reference(csr, fac, reg=0.3)
+    # dense check for a few rows
+    rows = torch.repeat_interleave(torch.arange(20), csr.row_counts())
+    for u in [0, 7, 19]:
+        mask = rows == u
+        q = fac[csr.indices.long()[mask]]
+        r = csr.values[mask]
+        n = int(mask.sum())
+        A_u = q.T @ q + 0.3 * n * torch.eye(k)
+        b_u = q.T @ r
+        assert torch.allclose(A[u], A_u, atol=1e-4)
+        assert torch.allclose(b[u], b_u, atol=1e-4)
+
+
+def test_cholesky_solve_matches_linalg():
+    g = torch.Generator().manual_seed(1)
+    B, k = 12, 16
+    M = torch.randn(B, k, k, generator=g)
+    A = M @ M.transpose(1, 2) + 0.5 * torch.eye(k)
+    b = torch.randn(B, k, generator=g)
+    x = R.cholesky_solve_reference(A, b)
+    assert torch.allclose(torch.linalg.solve(A, b), x, atol=1e-4)
+
+
+def test_cholesky_singular_rows_zero():
+    A = torch.zeros(2, 4, 4)
+    A[1] = torch.eye(4) * 2.0
+    b = torch.ones(2, 4)
+    x = R.cholesky_solve_reference(A, b)
+    assert torch.all(x[0] == 0)
+    assert torch.allclose(x[1], torch.full((4,), 0.5))
+
+
+def test_als_solve_side_decreases_residual():
+    csr = _rand_csr(rows=40, cols=25, nnz=500)
+    fac = torch.rand(25, 8)
+    p = R.als_solve_side_reference(csr, fac, reg=0.1)
+    assert p.shape == (40, 8)
+    assert torch.isfinite(p).all()
+
+
+def test_sgd_update_v1_semantics():
+    """v1 updates both vectors from the OLD copies (SGD.java:199-207)."""
+    p = torch.tensor([[1.0, 2.0]])
+    q = torch.tensor([[0.5, -1.0]])
+    r = torch.tensor([3.0])
+    lr, ureg, ireg = 0.1, 0.01, 0.02
+    pn, qn, err = R.sgd_update_reference(p, q, r, lr, ureg, ireg)
+    e = 3.0 - (1.0 * 0.5 + 2.0 * -1.0)
+    assert torch.allclose(err, torch.tensor([e]))
+    assert torch.allclose(pn, p + lr * (e * q - ureg * p))
+    # q update must see the OLD p (not pn)
+    assert torch.allclose(qn, q + lr * (e * p - ireg * q))
+
+
+def test_predict_dot():
+    U = torch.randn(5, 8)
+    V = torch.randn(6, 8)
+    out = R.predict_dot_reference(U, V, torch.tensor([0, 4]), torch.tensor([1, 5]))
+    assert torch.allclose(out[0], (U[0] * V[1]).sum())
+    assert torch.allclose(out[1], (U[4] * V[5]).sum())
+
+
+def test_sdca_reference_converges():
+    from flink_ms_amd.data.libsvm import LibSVMShape, synthetic_libsvm
+    csr, y = synthetic_libsvm(LibSVMShape(150, 20, 6), seed=5, separable=True)
+    alpha = torch.zeros(150)
+    v = torch.zeros(20)
+    o0 = R.hinge_objective(csr, y, v, 0.01)
+    for _ in range(5):
+        R.sdca_epoch_reference(csr, y, alpha, v, 0.01, 150)
+    o1 = R.hinge_objective(csr, y, v, 0.01)
+    assert o1 < o0
+    assert (alpha >= 0).all() and (alpha <= 1).all()

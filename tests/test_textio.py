@@ -1,0 +1,71 @@
+"""Golden tests for the text model formats (payload parity)."""
+
+import math
+
+import pytest
+
+from flink_ms_amd.utils import textio as t
+
+
+# Expected strings verified against Java Double.toString semantics.
+JAVA_DOUBLE_CASES = [
+    (1.0, "1.0"), (0.5, "0.5"), (100.0, "100.0"), (1e7, "1.0E7"),
+    (9999999.0, "9999999.0"), (0.001, "0.001"), (0.0009, "9.0E-4"),
+    (1e-4, "1.0E-4"), (-2.5, "-2.5"), (0.0, "0.0"), (-0.0, "-0.0"),
+    (123456789.0, "1.23456789E8"), (0.1, "0.1"),
+    (1 / 3, "0.3333333333333333"), (1.5e-10, "1.5E-10"),
+    (1.7976931348623157e308, "1.7976931348623157E308"),
+    (12345.678, "12345.678"), (2e7, "2.0E7"), (-1e-3, "-0.001"),
+    (42.0, "42.0"), (3.14159, "3.14159"), (-7.25e-5, "-7.25E-5"),
+]
+
+
+@pytest.mark.parametrize("x,expected", JAVA_DOUBLE_CASES)
+def test_java_double_format(x, expected):
+    assert t.java_double_to_string(x) == expected
+
+
+def test_java_double_specials():
+    assert t.java_double_to_string(float("nan")) == "NaN"
+    assert t.java_double_to_string(float("inf")) == "Infinity"
+    assert t.java_double_to_string(float("-inf")) == "-Infinity"
+
+
+def test_java_double_roundtrip():
+    import random
+    rng = random.Random(42)
+    for _ in range(2000):
+        x = rng.uniform(-1e9, 1e9) * 10 ** rng.randint(-12, 12)
+        s = t.java_double_to_string(x)
+        assert float(s.replace("E", "e")) == x, (x, s)
+
+
+def test_als_row_roundtrip():
+    row = t.als_factor_row(42, "U", [1.0, 0.5, -0.25])
+    assert row == "42,U,1.0;0.5;-0.25"
+    rid, kind, facs = t.parse_als_row(row)
+    assert (rid, kind, facs) == ("42", "U", [1.0, 0.5, -0.25])
+
+
+def test_mean_row():
+    row = t.als_factor_row(t.MEAN_ID, "I", [2.0])
+    assert row == "MEAN,I,2.0"
+    assert t.als_state_key("MEAN", "I") == "MEAN-I"
+    assert t.als_state_key(7, "U") == "7-U"
+
+
+def test_svm_rows():
+    assert t.svm_flat_row(1, 0.5) == "1,0.5"
+    assert t.parse_svm_flat_row("17,-2.0") == (17, -2.0)
+    # bucket = 1-based index / range (SVMImpl.scala:42)
+    assert t.svm_bucket_of(999, 1000) == 0
+    assert t.svm_bucket_of(1000, 1000) == 1
+    row = t.svm_range_row(2, [(2000, 1.5), (2001, -0.125)])
+    assert row == "2,2000:1.5;2001:-0.125"
+    b, pairs = t.parse_svm_range_row(row)
+    assert b == 2 and pairs == [(2000, 1.5), (2001, -0.125)]
+
+
+def test_latency_csv_rows():
+    assert t.als_latency_csv_row(1, 2, 3.5, 12) == "1,2,3.5,12"
+    assert t.svm_latency_csv_row(0, 5, -1.0, 3) == "0,5,-1.0,3"

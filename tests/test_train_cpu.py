@@ -1,0 +1,92 @@
+"""CPU end-to-end training tests (ALS + CoCoA-SVM + model file formats)."""
+
+import io
+
+import torch
+
+from flink_ms_amd.data.libsvm import LibSVMShape, synthetic_libsvm
+from flink_ms_amd.data.ratings import RatingsShape, synthetic_ratings
+from flink_ms_amd.models.als import ALSConfig, train_als
+from flink_ms_amd.models.mean_vector import mean_vector_rows
+from flink_ms_amd.models.mse import evaluate_mse
+from flink_ms_amd.models.svm import SVMConfig, SVMTrainer
+from flink_ms_amd.utils.textio import parse_als_row, parse_svm_range_row
+
+
+def test_als_converges_and_writes_model():
+    shape = RatingsShape(200, 100, 3000)
+    u, i, r = synthetic_ratings(shape, seed=1)
+    model, tr = train_als(u, i, r, shape.num_users, shape.num_items,
+                          ALSConfig(iterations=5, num_factors=16,
+                                    lambda_=0.1, dtype=torch.float32))
+    res = evaluate_mse(model.user_factors, model.item_factors, u, i, r)
+    assert res.mse < 1.0
+    uf, itf = io.StringIO(), io.StringIO()
+    model.write(uf, itf)
+    urows = uf.getvalue().strip().split("\n")
+    irows = itf.getvalue().strip().split("\n")
+    assert len(urows) == 200 and len(irows) == 100
+    rid, kind, facs = parse_als_row(urows[0])
+    assert rid == "0" and kind == "U" and len(facs) == 16
+    # mean-vector job consumes the same rows
+    mean_row = mean_vector_rows(urows, "U")
+    assert mean_row.startswith("MEAN,U,")
+    _, _, mfacs = parse_als_row(mean_row)
+    assert len(mfacs) == 16
+
+
+def test_als_mse_decreases_over_iterations():
+    shape = RatingsShape(150, 80, 2500)
+    u, i, r = synthetic_ratings(shape, seed=2)
+    mses = []
+    for iters in (1, 5):
+        model, _ = train_als(u, i, r, shape.num_users, shape.num_items,
+                             ALSConfig(iterations=iters, num_factors=8,
+                                       lambda_=0.2, dtype=torch.float32))
+        mses.append(evaluate_mse(model.user_factors, model.item_factors,
+                                 u, i, r).mse)
+    assert mses[1] <= mses[0] + 1e-6
+
+
+def test_svm_cocoa_converges_and_writes_model():
+    csr, y = synthetic_libsvm(LibSVMShape(400, 50, 10), seed=3, separable=True)
+    tr = SVMTrainer(SVMConfig(iterations=5, local_iterations=2,
+                              regularization=0.01))
+    tr.setup(csr, y)
+    o0 = tr.objective()
+    model = tr.fit()
+    o1 = tr.objective()
+    assert o1 < o0
+    flat = io.StringIO()
+    model.write_flat(flat)
+    rows = flat.getvalue().strip().split("\n")
+    assert len(rows) == 50
+    assert rows[0].startswith("1,")  # 1-based indices (SVMImpl.scala:33-35)
+    rp = io.StringIO()
+    model.write_range_partitioned(rp, range_size=16)
+    prow = rp.getvalue().strip().split("\n")[0]
+    bucket, pairs = parse_svm_range_row(prow)
+    assert bucket == 0 and pairs[0][0] == 1
+
+
+def test_generators_emit_reference_formats():
+    from flink_ms_amd.models.generator import generate_als_model, generate_svm_model
+    rows = list(generate_als_model(5, 3, 4, seed=1))
+    assert len(rows) == 8
+    rid, kind, facs = parse_als_row(rows[0])
+    assert rid == "1" and kind == "U" and len(facs) == 4
+    assert all(f >= 0 for f in facs)  # ratio of uniforms is nonnegative
+    srows = list(generate_svm_model(40, 10, seed=2))
+    assert len(srows) == 4
+    b, pairs = parse_svm_range_row(srows[1])
+    assert b == 1 and pairs[0][0] == 10  # 0-based keys from bucket*range
+    toks = srows[0].split(",", 1)[1].split(";")
+    assert any(tk.endswith(":0") for tk in toks)  # ~50% Int-typed zeros
+
+
+def test_bench_cpu_smoke():
+    import bench
+    rc = bench.main(["--device", "cpu", "--steps", "1", "--warmup", "0",
+                     "--rank", "8", "--ratings-per-gpu", "2000",
+                     "--users-per-gpu", "150", "--items", "80"])
+    assert rc == 0
