@@ -168,7 +168,8 @@ def predict_dot(U: torch.Tensor, V: torch.Tensor, u_idx: torch.Tensor,
     if U.is_cuda:
         ops = _require_hip()
         out = torch.empty(u_idx.numel(), dtype=torch.float32, device=U.device)
-        ops.predict_dot(U.contiguous(), V.contiguous(), u_idx.long(),
+        ops.predict_dot(U.to(torch.bfloat16).contiguous(),
+                        V.to(torch.bfloat16).contiguous(), u_idx.long(),
                         i_idx.long(), out, _stream())
         return out
     return reference.predict_dot_reference(U, V, u_idx, i_idx)

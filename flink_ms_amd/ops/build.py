@@ -25,7 +25,8 @@ CSRC = PKG_DIR / "ops" / "csrc"
 SO_PATH = PKG_DIR / "_hip_ops.so"
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
-KERNEL_SOURCES = ["als_kernels.hip", "svm_kernels.hip", "serve_kernels.hip"]
+KERNEL_SOURCES = ["als_kernels.hip", "svm_kernels.hip", "serve_kernels.hip",
+                  "debug_kernels.hip"]
 
 
 def _hipcc() -> str:
@@ -56,7 +57,8 @@ def build(verbose: bool = True, force: bool = False) -> Path:
     objdir.mkdir(exist_ok=True)
 
     srcs = [CSRC / s for s in KERNEL_SOURCES] + [CSRC / "bindings.cpp"]
-    deps = srcs + [CSRC / "common.hip.h", Path(__file__)]
+    deps = srcs + [CSRC / "common.hip.h", CSRC / "als_kernels_device.inc",
+            Path(__file__)]
     if not force and SO_PATH.exists():
         so_mtime = SO_PATH.stat().st_mtime
         if all(d.stat().st_mtime < so_mtime for d in deps):

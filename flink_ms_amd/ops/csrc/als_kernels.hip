@@ -31,265 +31,7 @@
 
 #include "common.hip.h"
 
-// ---------------------------------------------------------------- geometry
-
-template <int KT> struct Geo {
-    static constexpr int K = 16 * KT;
-    static constexpr int SP = ((K + 32 + 31) / 32) * 32;     // stage row stride (elements)
-    static constexpr int NA = KT * (KT + 1) / 2;             // upper-triangle tiles
-    static constexpr int TILES = NA + KT;                    // + EXT (b) tiles
-    static constexpr int SLOTS = (TILES + 3) / 4;            // acc tiles per wave
-    static constexpr int STAGE_BYTES = 32 * SP * 2;
-    static constexpr int A_BYTES = K * (K + 1) * 4;
-    static constexpr int SMEM = (STAGE_BYTES > A_BYTES + 8 * K)
-                                    ? STAGE_BYTES : A_BYTES + 8 * K;
-};
-
-DEV_INLINE unsigned stage_xor(int row) { return (row & 8) ? 32u : 0u; }
-
-// ------------------------------------------------------------------ stage
-// Gather the chunk's 32 factor rows (+ rating hi/lo block) into LDS.
-// nrem = ratings left in this entity (rows >= nrem are zero-filled).
-template <int KT>
-DEV_INLINE void stage_chunk(char* smem,
-                            const int* __restrict__ indices,
-                            const float* __restrict__ values,
-                            const unsigned short* __restrict__ factors,
-                            long long base, int nrem) {
-    constexpr int K = Geo<KT>::K, SP = Geo<KT>::SP;
-    constexpr int LPR = K / 8;              // 16B loads per factor row
-    const int tid = threadIdx.x;
-    // factor-row tasks
-    for (int t = tid; t < 32 * LPR; t += 256) {
-        const int row = t / LPR, seg = t % LPR;
-        uint4 v = {0, 0, 0, 0};
-        if (row < nrem) {
-            const long long col = indices[base + row];
-            v = *(const uint4*)(factors + col * (long long)K + seg * 8);
-        }
-        unsigned byte = (unsigned)(row * SP + seg * 8) * 2u ^ stage_xor(row);
-        *(uint4*)(smem + byte) = v;
-    }
-    // rating EXT block: columns K..K+15 = [hi, lo, 0 x14]
-    for (int t = tid; t < 64; t += 256) {
-        const int row = t >> 1, half = t & 1;
-        uint4 v = {0, 0, 0, 0};
-        if (half == 0 && row < nrem) {
-            const float r = values[base + row];
-            const unsigned short hi = f2bf(r);
-            const unsigned short lo = f2bf(r - bf2f(hi));
-            v.x = (unsigned)hi | ((unsigned)lo << 16);
-        }
-        unsigned byte = (unsigned)(row * SP + K + half * 8) * 2u ^ stage_xor(row);
-        *(uint4*)(smem + byte) = v;
-    }
-}
-
-// ------------------------------------------------------------- frag reads
-// A- and B-operand fragments of v_mfma_f32_16x16x32_bf16 share one layout
-// here: lane l holds G[(l>>4)*8 + j][tile*16 + (l&15)] for j=0..7.  The
-// Gramian sums over the staged-row (contraction) axis, so any consistent
-// lane->k mapping of the hardware yields the same A (both operands use the
-// same map); only the 16-row/16-col lane maps and the C/D layout
-// (col=lane&15, row=(lane>>4)*4+reg: cdna_hip_programming.md §3) must match
-// the hardware, which tests/test_gpu_mfma.py verifies.
-template <int KT>
-DEV_INLINE void read_frags(const char* smem, int lane, bf16x8* frag) {
-    constexpr int SP = Geo<KT>::SP;
-    const int g = lane >> 4, li = lane & 15;
-    const unsigned xorb = (g & 1) ? 32u : 0u;
-#pragma unroll
-    for (int t = 0; t <= KT; ++t) {
-        unsigned short e[8];
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-            unsigned byte = (unsigned)((8 * g + j) * SP + t * 16 + li) * 2u;
-            e[j] = *(const unsigned short*)(smem + (byte ^ xorb));
-        }
-        bf16x8 f;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) f[j] = (short)e[j];
-        frag[t] = f;
-    }
-}
-
-// --------------------------------------------------------------- MFMA loop
-
-template <int KT, int W, int S = 0>
-DEV_INLINE void mfma_tiles(const bf16x8* frag, f32x4* acc) {
-    if constexpr (S < Geo<KT>::SLOTS) {
-        constexpr int t = W + 4 * S;
-        if constexpr (t < Geo<KT>::TILES) {
-            constexpr int NA = Geo<KT>::NA;
-            constexpr int p = (t < NA) ? up_tile_p(t, KT) : (t - NA);
-            constexpr int q = (t < NA) ? up_tile_q((t < NA) ? t : 0, KT) : KT;
-            acc[S] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                frag[p], frag[q], acc[S], 0, 0, 0);
-        }
-        mfma_tiles<KT, W, S + 1>(frag, acc);
-    }
-}
-
-// Spill accumulators to the LDS A/b images (C/D map: row=(l>>4)*4+r, col=l&15).
-template <int KT, int W, int S = 0>
-DEV_INLINE void write_acc(const f32x4* acc, float* A, float* bhi, float* blo,
-                          int lane) {
-    if constexpr (S < Geo<KT>::SLOTS) {
-        constexpr int t = W + 4 * S;
-        if constexpr (t < Geo<KT>::TILES) {
-            constexpr int K = Geo<KT>::K;
-            constexpr int NA = Geo<KT>::NA;
-            constexpr int p = (t < NA) ? up_tile_p(t, KT) : (t - NA);
-            constexpr int q = (t < NA) ? up_tile_q((t < NA) ? t : 0, KT) : KT;
-            const int g = lane >> 4, li = lane & 15;
-#pragma unroll
-            for (int r = 0; r < 4; ++r) {
-                const int row = p * 16 + g * 4 + r;
-                if constexpr (q == KT) {       // EXT: b tile, cols 0/1 = hi/lo
-                    if (li == 0) bhi[row] = acc[S][r];
-                    else if (li == 1) blo[row] = acc[S][r];
-                } else {
-                    const int col = q * 16 + li;
-                    A[row * (K + 1) + col] = acc[S][r];
-                    if constexpr (p != q) A[col * (K + 1) + row] = acc[S][r];
-                }
-            }
-        }
-        write_acc<KT, W, S + 1>(acc, A, bhi, blo, lane);
-    }
-}
-
-// ------------------------------------------------------ in-LDS Cholesky/solve
-
-// Right-looking Cholesky of the [K][K+1] LDS image, lower triangle in place.
-// Thread (r = tid%K, s = tid/K) owns row r, column-slice s.
-template <int K>
-DEV_INLINE void cholesky_lds(float* A) {
-    constexpr int NSL = 256 / K;
-    const int tid = threadIdx.x;
-    const int r = tid % K, s = tid / K;
-    const bool owner = tid < NSL * K;
-    for (int j = 0; j < K; ++j) {
-        const float d = sqrtf(A[j * (K + 1) + j]);
-        const float dinv = 1.0f / d;
-        if (owner && s == 0 && r > j) A[r * (K + 1) + j] *= dinv;
-        __syncthreads();
-        if (owner && s == 0 && r == j) A[j * (K + 1) + j] = d;
-        if (owner && r > j) {
-            const float lrj = A[r * (K + 1) + j];
-            for (int c = j + 1 + s; c <= r; c += NSL)
-                A[r * (K + 1) + c] -= lrj * A[c * (K + 1) + j];
-        }
-        __syncthreads();
-    }
-}
-
-// Triangular solves L L^T x = b on wave 0; lane owns rows lane and lane+64.
-// Returns x in (x0, x1).
-template <int K>
-DEV_INLINE void solve_lds(const float* A, const float* b, int lane,
-                          float& x0, float& x1) {
-    const bool lv = lane < K;                       // low row valid
-    const bool hv = (K > 64) && (lane + 64 < K);    // high row valid
-    x0 = lv ? b[lane] : 0.0f;
-    x1 = hv ? b[lane + 64] : 0.0f;
-    const float id0 = lv ? 1.0f / A[lane * (K + 1) + lane] : 0.0f;
-    const float id1 = hv ? 1.0f / A[(lane + 64) * (K + 1) + lane + 64] : 0.0f;
-    // forward: L y = b
-    for (int j = 0; j < K; ++j) {
-        const bool hi = j >= 64;
-        const float yj = __shfl(hi ? x1 : x0, j & 63, WAVE)
-                       * __shfl(hi ? id1 : id0, j & 63, WAVE);
-        if (!hi) {
-            if (lane == j) x0 = yj;
-            else if (lane > j && lv) x0 -= A[lane * (K + 1) + j] * yj;
-            if (hv) x1 -= A[(lane + 64) * (K + 1) + j] * yj;
-        } else {
-            if (lane + 64 == j) x1 = yj;
-            else if (lane + 64 > j && hv) x1 -= A[(lane + 64) * (K + 1) + j] * yj;
-        }
-    }
-    // backward: L^T x = y  (L^T[r][j] = A[j][r])
-    for (int j = K - 1; j >= 0; --j) {
-        const bool hi = j >= 64;
-        const float xj = __shfl(hi ? x1 : x0, j & 63, WAVE)
-                       * __shfl(hi ? id1 : id0, j & 63, WAVE);
-        if (hi) {
-            if (lane + 64 == j) x1 = xj;
-            else if (lane + 64 < j && hv) x1 -= A[j * (K + 1) + lane + 64] * xj;
-            if (lv) x0 -= A[j * (K + 1) + lane] * xj;
-        } else {
-            if (lane == j) x0 = xj;
-            else if (lane < j && lv) x0 -= A[j * (K + 1) + lane] * xj;
-        }
-    }
-}
-
-// ------------------------------------------------- Gramian accumulation core
-// Runs the chunked stage->frag->MFMA loop and leaves A (mirrored, with
-// lambda*n*I) + combined b in LDS.  Returns n (ratings of this entity).
-template <int KT>
-DEV_INLINE int gramian_to_lds(char* smem,
-                              const long long* __restrict__ indptr,
-                              const int* __restrict__ indices,
-                              const float* __restrict__ values,
-                              const unsigned short* __restrict__ factors,
-                              long long row, float reg) {
-    constexpr int K = Geo<KT>::K;
-    const int tid = threadIdx.x, lane = tid & 63, w = tid >> 6;
-    const long long p0 = indptr[row];
-    const int n = (int)(indptr[row + 1] - p0);
-    if (n == 0) return 0;
-
-    f32x4 acc[Geo<KT>::SLOTS];
-#pragma unroll
-    for (int s = 0; s < Geo<KT>::SLOTS; ++s) acc[s] = f32x4{0, 0, 0, 0};
-
-    const int nchunks = (n + 31) >> 5;
-    for (int ch = 0; ch < nchunks; ++ch) {
-        stage_chunk<KT>(smem, indices, values, factors, p0 + ch * 32,
-                        n - ch * 32);
-        __syncthreads();
-        bf16x8 frag[KT + 1];
-        read_frags<KT>(smem, lane, frag);
-        switch (w) {
-            case 0: mfma_tiles<KT, 0>(frag, acc); break;
-            case 1: mfma_tiles<KT, 1>(frag, acc); break;
-            case 2: mfma_tiles<KT, 2>(frag, acc); break;
-            default: mfma_tiles<KT, 3>(frag, acc); break;
-        }
-        __syncthreads();
-    }
-
-    float* A = (float*)smem;
-    float* bhi = A + K * (K + 1);
-    float* blo = bhi + K;
-    switch (w) {
-        case 0: write_acc<KT, 0>(acc, A, bhi, blo, lane); break;
-        case 1: write_acc<KT, 1>(acc, A, bhi, blo, lane); break;
-        case 2: write_acc<KT, 2>(acc, A, bhi, blo, lane); break;
-        default: write_acc<KT, 3>(acc, A, bhi, blo, lane); break;
-    }
-    __syncthreads();
-    if (tid < K) {
-        bhi[tid] += blo[tid];                       // combined b
-        float dd = A[tid * (K + 1) + tid] + reg * (float)n;
-        if (dd <= 0.0f) dd = 1.0f;                  // degenerate guard
-        A[tid * (K + 1) + tid] = dd;
-    }
-    __syncthreads();
-    return n;
-}
-
-template <int KT>
-DEV_INLINE void write_zero_row(float* out_f32, unsigned short* out_bf16,
-                               long long row) {
-    constexpr int K = Geo<KT>::K;
-    for (int c = threadIdx.x; c < K; c += 256) {
-        out_f32[row * K + c] = 0.0f;
-        if (out_bf16) out_bf16[row * K + c] = 0;
-    }
-}
+#include "als_kernels_device.inc"
 
 // -------------------------------------------------------------- kernels
 
@@ -447,19 +189,6 @@ __global__ void k_mfma_probe_bf16(const unsigned short* __restrict__ Xt,
 }
 
 // -------------------------------------------------------------- launchers
-
-#define DISPATCH_KT(k, expr)                                                  \
-    switch ((k) / 16) {                                                       \
-        case 1: { constexpr int KT = 1; expr; break; }                        \
-        case 2: { constexpr int KT = 2; expr; break; }                        \
-        case 3: { constexpr int KT = 3; expr; break; }                        \
-        case 4: { constexpr int KT = 4; expr; break; }                        \
-        case 5: { constexpr int KT = 5; expr; break; }                        \
-        case 6: { constexpr int KT = 6; expr; break; }                        \
-        case 7: { constexpr int KT = 7; expr; break; }                        \
-        case 8: { constexpr int KT = 8; expr; break; }                        \
-        default: return hipErrorInvalidValue;                                 \
-    }
 
 extern "C" hipError_t fma_als_solve_fused(
     int k, const long long* indptr, const int* indices, const float* values,

@@ -39,6 +39,16 @@ int fma_sgd_update(unsigned short* U, unsigned short* V,
 int fma_mfma_probe_f32(const float* A, const float* B, float* D, void* stream);
 int fma_mfma_probe_bf16(const unsigned short* Xt, const unsigned short* Yt,
                         float* C, void* stream);
+int fma_dbg_stage_dump(int k, const int64_t* indptr, const int* indices,
+                       const float* values, const unsigned short* factors,
+                       unsigned short* out, void* stream);
+int fma_dbg_frag_dump(int k, const int64_t* indptr, const int* indices,
+                      const float* values, const unsigned short* factors,
+                      unsigned short* out, void* stream);
+int fma_dbg_gramian(int k, const int64_t* indptr, const int* indices,
+                    const float* values, const unsigned short* factors,
+                    float* A_out, float* bhi_out, float* blo_out,
+                    int64_t nrows, float reg, void* stream);
 const char* fma_err_str(int err);
 }
 
@@ -222,6 +232,42 @@ void mfma_probe_bf16(torch::Tensor Xt, torch::Tensor Yt, torch::Tensor C,
               "mfma_probe_bf16");
 }
 
+void dbg_stage_dump(torch::Tensor indptr, torch::Tensor indices,
+                    torch::Tensor values, torch::Tensor factors,
+                    torch::Tensor out, int64_t stream) {
+    check_hip(fma_dbg_stage_dump((int)factors.size(1),
+                                 indptr.data_ptr<int64_t>(),
+                                 indices.data_ptr<int>(),
+                                 values.data_ptr<float>(), bf16_ptr(factors),
+                                 bf16_ptr_mut(out), (void*)stream),
+              "dbg_stage_dump");
+}
+
+void dbg_frag_dump(torch::Tensor indptr, torch::Tensor indices,
+                   torch::Tensor values, torch::Tensor factors,
+                   torch::Tensor out, int64_t stream) {
+    check_hip(fma_dbg_frag_dump((int)factors.size(1),
+                                indptr.data_ptr<int64_t>(),
+                                indices.data_ptr<int>(),
+                                values.data_ptr<float>(), bf16_ptr(factors),
+                                bf16_ptr_mut(out), (void*)stream),
+              "dbg_frag_dump");
+}
+
+void dbg_gramian(torch::Tensor indptr, torch::Tensor indices,
+                 torch::Tensor values, torch::Tensor factors,
+                 torch::Tensor A_out, torch::Tensor bhi, torch::Tensor blo,
+                 double reg, int64_t stream) {
+    check_hip(fma_dbg_gramian((int)factors.size(1),
+                              indptr.data_ptr<int64_t>(),
+                              indices.data_ptr<int>(),
+                              values.data_ptr<float>(), bf16_ptr(factors),
+                              A_out.data_ptr<float>(), bhi.data_ptr<float>(),
+                              blo.data_ptr<float>(), indptr.size(0) - 1,
+                              (float)reg, (void*)stream),
+              "dbg_gramian");
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -233,6 +279,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("svm_margins", &svm_margins);
     m.def("predict_dot", &predict_dot);
     m.def("sgd_update", &sgd_update);
+    m.def("dbg_stage_dump", &dbg_stage_dump);
+    m.def("dbg_frag_dump", &dbg_frag_dump);
+    m.def("dbg_gramian", &dbg_gramian);
     m.def("mfma_probe_f32", &mfma_probe_f32);
     m.def("mfma_probe_bf16", &mfma_probe_bf16);
 }

@@ -83,8 +83,12 @@ extern "C" hipError_t fma_sdca_pass(
     const float* y, const float* norms_sq, const int* perm, float* alpha,
     float* v, long long nrows, float scale, hipStream_t stream) {
     if (nrows <= 0) return hipErrorInvalidValue;
-    long long waves = (nrows + 3) / 4;
-    unsigned grid = (unsigned)(waves < 2048 ? waves : 2048);
+    // keep >=16 samples sequential per wave: full-width hogwild degenerates
+    // to a synchronous full-batch step and stalls dual convergence
+    long long waves = nrows / 16;
+    if (waves < 4) waves = 4;
+    if (waves > 8192) waves = 8192;
+    unsigned grid = (unsigned)((waves + 3) / 4);
     k_sdca_pass<<<dim3(grid), dim3(256), 0, stream>>>(
         indptr, indices, values, y, norms_sq, perm, alpha, v, nrows, scale);
     return hipGetLastError();
