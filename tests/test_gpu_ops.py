@@ -136,15 +136,22 @@ def test_sgd_update_parity(gpu):
 
 
 def test_als_trainer_gpu_converges(gpu):
-    from flink_ms_amd.data.ratings import RatingsShape, synthetic_ratings
+    """Planted low-rank ratings: the bf16 fused-solve path must recover the
+    structure (random-noise ratings are information-free and only bound by
+    the ridge; a planted matrix is the real convergence check)."""
     from flink_ms_amd.models.als import ALSConfig, ALSTrainer
     from flink_ms_amd.models.mse import evaluate_mse
-    shape = RatingsShape(3000, 1000, 100_000)
-    u, i, r = synthetic_ratings(shape, seed=11)
-    tr = ALSTrainer(ALSConfig(iterations=3, num_factors=64, lambda_=0.3))
+    g = torch.Generator().manual_seed(11)
+    U0 = torch.randn(3000, 8, generator=g) * 0.5
+    V0 = torch.randn(1000, 8, generator=g) * 0.5
+    u = torch.randint(0, 3000, (100_000,), generator=g)
+    i = torch.randint(0, 1000, (100_000,), generator=g)
+    r = (U0[u] * V0[i]).sum(dim=1)
+    tr = ALSTrainer(ALSConfig(iterations=5, num_factors=16, lambda_=0.02))
     tr.ctx.device = gpu
-    tr.setup(u.long(), i.long(), r, shape.num_users, shape.num_items)
+    tr.setup(u.long(), i.long(), r, 3000, 1000)
     tr.fit()
     m = tr.model()
     res = evaluate_mse(m.user_factors.to(gpu), m.item_factors.to(gpu), u, i, r)
-    assert res.mse < 1.2, f"GPU ALS MSE {res.mse}"
+    var = float(r.var())
+    assert res.mse < 0.1 * var, f"GPU ALS MSE {res.mse} vs rating var {var}"
