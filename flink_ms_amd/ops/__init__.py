@@ -71,21 +71,38 @@ def als_solve_side(
     reg: float,
     out_bf16: Optional[torch.Tensor] = None,
     row_order: Optional[torch.Tensor] = None,
+    fused: bool = False,
 ) -> torch.Tensor:
     """One ALS half-iteration: solve every row entity of ``csr`` against the
     opposite side's factors.  Returns fp32 [num_rows, k]; optionally also
-    writes the bf16 image for the next half-iteration in the same kernel."""
+    writes the bf16 image for the next half-iteration.
+
+    Two GPU paths: the default modular path (MFMA Gramian kernel -> batched
+    wave-per-entity LDL solve; best measured occupancy at k <= 64) and the
+    single-launch fused kernel (``fused=True``; avoids the A round-trip
+    through HBM but serializes the solve at Gramian occupancy).
+    """
     if other_factors.is_cuda:
         ops = _require_hip()
         fac = _pad_k(other_factors.to(torch.bfloat16).contiguous())
         k = fac.shape[1]
-        out = torch.empty(csr.num_rows, k, dtype=torch.float32,
-                          device=fac.device)
         ob = out_bf16 if out_bf16 is not None else _empty(fac.device)
         ro = row_order if row_order is not None else _empty(fac.device)
-        ops.als_solve_fused(csr.indptr, csr.indices, csr.values, fac,
-                            out, ob, ro, float(reg), _stream())
         korig = other_factors.shape[1]
+        if fused or k > 64:
+            out = torch.empty(csr.num_rows, k, dtype=torch.float32,
+                              device=fac.device)
+            ops.als_solve_fused(csr.indptr, csr.indices, csr.values, fac,
+                                out, ob, ro, float(reg), _stream())
+        else:
+            A = torch.empty(csr.num_rows, k, k, dtype=torch.float32,
+                            device=fac.device)
+            b = torch.empty(csr.num_rows, k, dtype=torch.float32,
+                            device=fac.device)
+            ops.gramian(csr.indptr, csr.indices, csr.values, fac, A, b,
+                        float(reg), _stream())
+            out = torch.empty_like(b)
+            ops.ldl_solve_wave(A, b, out, ob, _stream())
         return out[:, :korig] if korig != k else out
     return reference.als_solve_side_reference(csr, other_factors, reg)
 

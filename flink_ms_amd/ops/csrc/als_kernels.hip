@@ -106,12 +106,13 @@ __global__ void k_gramian(const long long* __restrict__ indptr,
 }
 
 // Standalone K2: batched SPD solve from dense global A/b.
+// phases bitmask (debug/ablation): 1 = eliminate, 2 = solve
 template <int KT>
 __launch_bounds__(256)
 __global__ void k_cholesky_solve(const float* __restrict__ A_in,  // [n][K][K]
                                  const float* __restrict__ b_in,  // [n][K]
                                  float* __restrict__ x_out,       // [n][K]
-                                 long long nrows) {
+                                 long long nrows, int phases) {
     constexpr int K = Geo<KT>::K;
     __shared__ __align__(16) char smem[Geo<KT>::SMEM];
     const long long row = blockIdx.x;
@@ -123,10 +124,10 @@ __global__ void k_cholesky_solve(const float* __restrict__ A_in,  // [n][K][K]
         A[(i / K) * (K + 1) + (i % K)] = A_in[row * K * K + i];
     for (int c = tid; c < K; c += 256) b[c] = b_in[row * K + c];
     __syncthreads();
-    cholesky_lds<K>(A);
+    if (phases & 1) cholesky_lds<K>(A);
     if (tid < 64) {
-        float x0, x1;
-        solve_lds<K>(A, b, tid, x0, x1);
+        float x0 = b[tid & 63], x1 = 0.0f;
+        if (phases & 2) solve_lds<K>(A, b, tid, x0, x1);
         if (tid < K) x_out[row * K + tid] = x0;
         if (K > 64 && tid + 64 < K) x_out[row * K + tid + 64] = x1;
     }
@@ -213,13 +214,101 @@ extern "C" hipError_t fma_gramian(
     return hipGetLastError();
 }
 
-extern "C" hipError_t fma_cholesky_solve(
+extern "C" hipError_t fma_cholesky_solve_ph(
     int k, const float* A_in, const float* b_in, float* x_out,
-    long long nrows, hipStream_t stream) {
+    long long nrows, int phases, hipStream_t stream) {
     if (k % 16 || k < 16 || k > 128 || nrows <= 0) return hipErrorInvalidValue;
     dim3 grid((unsigned)nrows), block(256);
     DISPATCH_KT(k, (k_cholesky_solve<KT><<<grid, block, 0, stream>>>(
-        A_in, b_in, x_out, nrows)));
+        A_in, b_in, x_out, nrows, phases)));
+    return hipGetLastError();
+}
+
+extern "C" hipError_t fma_cholesky_solve(
+    int k, const float* A_in, const float* b_in, float* x_out,
+    long long nrows, hipStream_t stream) {
+    return fma_cholesky_solve_ph(k, A_in, b_in, x_out, nrows, 3, stream);
+}
+
+
+// -------- wave-per-entity LDL solver (k <= 64) --------
+// The block-per-entity factorization is issue-bound: ~100 instructions of
+// work between 2x64 barriers.  For k <= 64 a single wave owns one entity's
+// whole [K][K+1] LDS image instead: no barriers (wave-internal LDS ordering
+// suffices), 4 independent entities per block, solve fully in registers.
+template <int KT>
+__launch_bounds__(256)
+__global__ void k_ldl_solve_wave(const float* __restrict__ A_in,
+                                 const float* __restrict__ b_in,
+                                 float* __restrict__ x_out,
+                                 unsigned short* __restrict__ x_bf16,
+                                 long long nrows) {
+    constexpr int K = KT * 16;
+    static_assert(K <= 64, "wave solver handles k <= 64");
+    constexpr int LDA = K + 1;
+    __shared__ __align__(16) float As[4][K * LDA];
+    const int w = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
+    const int lane = threadIdx.x & 63;
+    const long long e = (long long)blockIdx.x * 4 + w;
+    if (e >= nrows) return;
+    float* A = As[w];
+    const float* src = A_in + e * (long long)(K * K);
+    for (int i = lane; i < K * K; i += 64)
+        A[(i / K) * LDA + (i % K)] = src[i];
+    float x0 = (lane < K) ? b_in[e * K + lane] : 0.0f;
+    // in-place LDL^T elimination (column j stays raw; see cholesky_lds)
+    for (int j = 0; j < K - 1; ++j) {
+        const float dj = A[j * LDA + j];
+        const float dinv = dj > 0.0f ? 1.0f / dj : 0.0f;
+        if (lane > j && lane < K) {
+            const float lscl = A[lane * LDA + j] * dinv;
+            float* rowp = A + lane * LDA;
+            const float* colp = A + j;
+            for (int c0 = j + 1; c0 <= lane; c0 += 4) {
+                float colv[4], rowv[4];
+#pragma unroll
+                for (int m = 0; m < 4; ++m)
+                    if (c0 + m <= lane) {
+                        colv[m] = colp[(c0 + m) * LDA];
+                        rowv[m] = rowp[c0 + m];
+                    }
+#pragma unroll
+                for (int m = 0; m < 4; ++m)
+                    if (c0 + m <= lane)
+                        rowp[c0 + m] = rowv[m] - lscl * colv[m];
+            }
+        }
+    }
+    // solve (I+Ls) y = b ; w = y/D ; (I+Ls^T) x = w, all in registers
+    const float d0 = (lane < K) ? A[lane * LDA + lane] : 1.0f;
+    const float id0 = d0 > 0.0f ? 1.0f / d0 : 0.0f;
+    for (int j = 0; j < K - 1; ++j) {
+        const float zj = __shfl(x0, j, WAVE) * __shfl(id0, j, WAVE);
+        if (lane > j && lane < K) x0 -= A[lane * LDA + j] * zj;
+    }
+    x0 *= id0;
+    for (int c = K - 1; c >= 1; --c) {
+        const float xc = __shfl(x0, c, WAVE);
+        if (lane < c) x0 -= A[c * LDA + lane] * id0 * xc;
+    }
+    if (lane < K) {
+        x_out[e * K + lane] = x0;
+        if (x_bf16) x_bf16[e * K + lane] = f2bf(x0);
+    }
+}
+
+extern "C" hipError_t fma_ldl_solve_wave(
+    int k, const float* A_in, const float* b_in, float* x_out,
+    unsigned short* x_bf16, long long nrows, hipStream_t stream) {
+    if (k % 16 || k < 16 || k > 64 || nrows <= 0) return hipErrorInvalidValue;
+    dim3 grid((unsigned)((nrows + 3) / 4)), block(256);
+    switch (k / 16) {
+        case 1: k_ldl_solve_wave<1><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, nrows); break;
+        case 2: k_ldl_solve_wave<2><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, nrows); break;
+        case 3: k_ldl_solve_wave<3><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, nrows); break;
+        case 4: k_ldl_solve_wave<4><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, nrows); break;
+        default: return hipErrorInvalidValue;
+    }
     return hipGetLastError();
 }
 

@@ -22,6 +22,9 @@ int fma_gramian(int k, const int64_t* indptr, const int* indices,
                 void* stream);
 int fma_cholesky_solve(int k, const float* A_in, const float* b_in,
                        float* x_out, int64_t nrows, void* stream);
+int fma_cholesky_solve_ph(int k, const float* A_in, const float* b_in,
+                          float* x_out, int64_t nrows, int phases,
+                          void* stream);
 int fma_sdca_pass(const int64_t* indptr, const int* indices,
                   const float* values, const float* y, const float* norms_sq,
                   const int* perm, float* alpha, float* v, int64_t nrows,
@@ -36,6 +39,9 @@ int fma_sgd_update(unsigned short* U, unsigned short* V,
                    const int64_t* u_idx, const int64_t* i_idx,
                    const float* r, float* err_out, int64_t nq, int k,
                    float lr, float user_reg, float item_reg, void* stream);
+int fma_ldl_solve_wave(int k, const float* A_in, const float* b_in,
+                       float* x_out, unsigned short* x_bf16, int64_t nrows,
+                       void* stream);
 int fma_mfma_probe_f32(const float* A, const float* B, float* D, void* stream);
 int fma_mfma_probe_bf16(const unsigned short* Xt, const unsigned short* Yt,
                         float* C, void* stream);
@@ -121,6 +127,23 @@ void gramian(torch::Tensor indptr, torch::Tensor indices, torch::Tensor values,
                           b_out.data_ptr<float>(), nrows, (float)reg,
                           (void*)stream),
               "gramian");
+}
+
+void cholesky_solve_ph(torch::Tensor A, torch::Tensor b, torch::Tensor x,
+                       int64_t phases, int64_t stream) {
+    check_hip(fma_cholesky_solve_ph((int)A.size(1), A.data_ptr<float>(),
+                                    b.data_ptr<float>(), x.data_ptr<float>(),
+                                    A.size(0), (int)phases, (void*)stream),
+              "cholesky_solve_ph");
+}
+
+void ldl_solve_wave(torch::Tensor A, torch::Tensor b, torch::Tensor x,
+                    torch::Tensor x_bf16, int64_t stream) {
+    unsigned short* xb = x_bf16.numel() > 0 ? bf16_ptr_mut(x_bf16) : nullptr;
+    check_hip(fma_ldl_solve_wave((int)A.size(1), A.data_ptr<float>(),
+                                 b.data_ptr<float>(), x.data_ptr<float>(),
+                                 xb, A.size(0), (void*)stream),
+              "ldl_solve_wave");
 }
 
 void cholesky_solve(torch::Tensor A, torch::Tensor b, torch::Tensor x,
@@ -275,6 +298,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("als_solve_fused", &als_solve_fused);
     m.def("gramian", &gramian);
     m.def("cholesky_solve", &cholesky_solve);
+    m.def("cholesky_solve_ph", &cholesky_solve_ph);
+    m.def("ldl_solve_wave", &ldl_solve_wave);
     m.def("sdca_pass", &sdca_pass);
     m.def("svm_margins", &svm_margins);
     m.def("predict_dot", &predict_dot);

@@ -155,3 +155,20 @@ def test_als_trainer_gpu_converges(gpu):
     res = evaluate_mse(m.user_factors.to(gpu), m.item_factors.to(gpu), u, i, r)
     var = float(r.var())
     assert res.mse < 0.1 * var, f"GPU ALS MSE {res.mse} vs rating var {var}"
+
+
+@pytest.mark.parametrize("k", [16, 32, 64])
+def test_ldl_wave_solver_parity(gpu, k):
+    import flink_ms_amd._hip_ops as hip
+    g = torch.Generator().manual_seed(21)
+    B = 257  # odd batch exercises the tail wave
+    M = torch.randn(B, k, k, generator=g) * 0.3
+    A = (M @ M.transpose(1, 2) + 1.5 * torch.eye(k)).contiguous().to(gpu)
+    b = torch.randn(B, k, generator=g).to(gpu)
+    x = torch.empty_like(b)
+    xb = torch.empty(B, k, dtype=torch.bfloat16, device=gpu)
+    hip.ldl_solve_wave(A, b, x, xb, torch.cuda.current_stream().cuda_stream)
+    x_ref = R.cholesky_solve_reference(A.cpu(), b.cpu()).to(gpu)
+    torch.cuda.synchronize()
+    assert torch.allclose(x, x_ref, atol=1e-3, rtol=1e-3)
+    assert torch.allclose(xb.to(torch.float32), x, atol=1e-1, rtol=2e-2)
