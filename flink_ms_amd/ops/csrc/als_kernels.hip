@@ -256,26 +256,47 @@ __global__ void k_ldl_solve_wave(const float* __restrict__ A_in,
     for (int i = lane; i < K * K; i += 64)
         A[(i / K) * LDA + (i % K)] = src[i];
     float x0 = (lane < K) ? b_in[e * K + lane] : 0.0f;
-    // in-place LDL^T elimination (column j stays raw; see cholesky_lds)
-    for (int j = 0; j < K - 1; ++j) {
-        const float dj = A[j * LDA + j];
-        const float dinv = dj > 0.0f ? 1.0f / dj : 0.0f;
-        if (lane > j && lane < K) {
-            const float lscl = A[lane * LDA + j] * dinv;
-            float* rowp = A + lane * LDA;
-            const float* colp = A + j;
-            for (int c0 = j + 1; c0 <= lane; c0 += 4) {
-                float colv[4], rowv[4];
+    // In-place LDL^T elimination, rank-4 panels; column values stay raw.
+    // Lanes run in wave lockstep, so panel-internal cross-lane dependencies
+    // (lane r reads lane c's row element) are ordered by program order.
+    // Cells (r,c) with c > r are never-read upper-triangle scratch, so the
+    // update runs guard-free over the full trailing range (per-element
+    // guards make hipcc branch around every load: cdna_hip_programming.md
+    // §5 ".s-level traps" (c)).  The joint update amortizes one row
+    // read/write over the panel's 4 columns.
+    if (lane < K) {
+        float* rowp = A + lane * LDA;
+        for (int p = 0; p < K; p += 4) {
+            float lscl[4];
+            // factor the 4-column panel (small triangle, lockstep)
 #pragma unroll
-                for (int m = 0; m < 4; ++m)
-                    if (c0 + m <= lane) {
-                        colv[m] = colp[(c0 + m) * LDA];
-                        rowv[m] = rowp[c0 + m];
-                    }
+            for (int jj = 0; jj < 4; ++jj) {
+                const int j = p + jj;
+                const float dj = A[j * LDA + j];
+                const float dinv = dj > 0.0f ? 1.0f / dj : 0.0f;
+                lscl[jj] = rowp[j] * dinv;
 #pragma unroll
-                for (int m = 0; m < 4; ++m)
-                    if (c0 + m <= lane)
-                        rowp[c0 + m] = rowv[m] - lscl * colv[m];
+                for (int cc = jj + 1; cc < 4; ++cc)
+                    rowp[p + cc] -= lscl[jj] * A[(p + cc) * LDA + j];
+            }
+            // joint trailing update: rows x remaining columns, 4 deep
+            for (int c0 = p + 4; c0 < K; c0 += 4) {
+                float rowv[4], colv[4][4];
+#pragma unroll
+                for (int m = 0; m < 4; ++m) {
+                    rowv[m] = rowp[c0 + m];
+#pragma unroll
+                    for (int jj = 0; jj < 4; ++jj)
+                        colv[m][jj] = A[(c0 + m) * LDA + p + jj];
+                }
+#pragma unroll
+                for (int m = 0; m < 4; ++m) {
+                    float v = rowv[m];
+#pragma unroll
+                    for (int jj = 0; jj < 4; ++jj)
+                        v -= lscl[jj] * colv[m][jj];
+                    rowp[c0 + m] = v;
+                }
             }
         }
     }
