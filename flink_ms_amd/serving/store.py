@@ -1,0 +1,297 @@
+"""In-process model stores: the queryable-state layer rebuilt.
+
+The reference serves models as Flink keyed state fed from Kafka
+(als-ms/.../qs/ALSKafkaConsumer.java:67-92, svm-ms/.../qs/SVMKafkaConsumer.java)
+and queried through the KvState protocol (QueryClientHelper.queryState ->
+``Optional``).  Here the state lives in ONE process:
+
+- the authoritative state is the text payload, exactly as ingested
+  (key -> ``(key, payload)`` Tuple2 parity; ALSKafkaConsumer.java:79-82
+  stores ``tokens[2]`` verbatim), so point queries return byte-identical
+  values;
+- alongside it, a device-resident bf16 factor mirror feeds the batched
+  GPU paths (load generators, MSE, online-SGD batches) through the K4/K5
+  serving kernels.
+
+Model hot-swap == re-ingesting rows (last writer wins per key), the same
+contract as re-publishing to the Kafka topic.
+"""
+
+from __future__ import annotations
+
+import threading
+from typing import Dict, Iterable, List, Optional, Tuple
+
+import torch
+
+from ..utils.textio import (
+    MEAN_ID,
+    als_state_key,
+    java_double_to_string,
+    parse_als_row,
+)
+
+ALS_STATE_NAME = "ALS_MODEL"   # ALSKafkaConsumer.java:91
+SVM_STATE_NAME = "SVM_MODEL"   # SVMKafkaConsumer.java:91
+
+
+class ALSModelStore:
+    """Keyed store ``"<id>-U" / "<id>-I" / "MEAN-U" / "MEAN-I"`` -> factors."""
+
+    def __init__(self, device: Optional[torch.device] = None):
+        self.device = device or (
+            torch.device("cuda:0") if torch.cuda.is_available()
+            else torch.device("cpu"))
+        self._payload: Dict[str, str] = {}       # key -> factor string
+        self._vec: Dict[str, List[float]] = {}   # parsed cache (fp64 path)
+        self._lock = threading.RLock()
+        # device mirror (built lazily; rebuilt on rank change)
+        self._rows: Dict[str, int] = {}          # key -> row in mirror
+        self._mirror: Optional[torch.Tensor] = None  # [cap, k] bf16
+        self._mirror_len = 0
+        self._k: Optional[int] = None
+
+    # ------------------------------------------------------------ ingest
+
+    def ingest_row(self, row: str) -> str:
+        """Consume one model row ``<id>,<U|I>,<f;f;...>``; returns the state
+        key (ALSKafkaConsumer map semantics: key = tokens[0]+"-"+tokens[1])."""
+        entity_id, kind, factors = parse_als_row(row)
+        key = als_state_key(entity_id, kind)
+        payload = row.strip().split(",", 2)[2]
+        with self._lock:
+            self._payload[key] = payload
+            self._vec[key] = factors
+            if self._k is None:
+                self._k = len(factors)
+            self._mirror_put(key, factors)
+        return key
+
+    def ingest(self, rows: Iterable[str]) -> int:
+        n = 0
+        for row in rows:
+            row = row.strip()
+            if not row:
+                continue
+            self.ingest_row(row)
+            n += 1
+        return n
+
+    def _mirror_put(self, key: str, factors: List[float]) -> None:
+        k = self._k
+        if k is None or len(factors) != k:
+            return
+        if self._mirror is None or self._mirror.shape[1] != k:
+            self._mirror = torch.zeros(1024, k, dtype=torch.bfloat16,
+                                       device=self.device)
+            self._rows.clear()
+            self._mirror_len = 0
+        row = self._rows.get(key)
+        if row is None:
+            if self._mirror_len == self._mirror.shape[0]:
+                bigger = torch.zeros(self._mirror.shape[0] * 2, k,
+                                     dtype=torch.bfloat16, device=self.device)
+                bigger[: self._mirror_len] = self._mirror
+                self._mirror = bigger
+            row = self._mirror_len
+            self._mirror_len += 1
+            self._rows[key] = row
+        self._mirror[row] = torch.tensor(factors, dtype=torch.float32
+                                         ).to(torch.bfloat16)
+
+    # ------------------------------------------------------------- query
+
+    def query(self, key: str) -> Optional[Tuple[str, str]]:
+        """Point lookup; ``None`` == Optional.empty (unknown key,
+        QueryClientHelper.java:135-137)."""
+        with self._lock:
+            payload = self._payload.get(key)
+        return None if payload is None else (key, payload)
+
+    def get_vector(self, key: str) -> Optional[List[float]]:
+        with self._lock:
+            return self._vec.get(key)
+
+    def predict(self, user_id: str, item_id: str) -> Optional[float]:
+        """``dot(U[u], V[i])`` in fp64 from the payloads — bit-matches the
+        reference client math (ALSPredict.java:74-83)."""
+        u = self.get_vector(als_state_key(user_id, "U"))
+        v = self.get_vector(als_state_key(item_id, "I"))
+        if u is None or v is None:
+            return None
+        return float(sum(a * b for a, b in zip(u, v)))
+
+    def predict_batch(self, user_ids: List[str], item_ids: List[str]
+                      ) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Batched predictions via the K5 kernel on the device mirror.
+        Returns (predictions fp32, found mask)."""
+        from .. import ops
+        with self._lock:
+            u_rows = [self._rows.get(als_state_key(u, "U"), -1)
+                      for u in user_ids]
+            i_rows = [self._rows.get(als_state_key(i, "I"), -1)
+                      for i in item_ids]
+            mirror = self._mirror
+        u_t = torch.tensor(u_rows, dtype=torch.int64)
+        i_t = torch.tensor(i_rows, dtype=torch.int64)
+        ok = (u_t >= 0) & (i_t >= 0)
+        preds = torch.zeros(len(user_ids), dtype=torch.float32)
+        if mirror is not None and int(ok.sum()) > 0:
+            sel_u = u_t[ok].to(mirror.device)
+            sel_i = i_t[ok].to(mirror.device)
+            p = ops.predict_dot(mirror, mirror, sel_u, sel_i)
+            preds[ok] = p.cpu()
+        return preds, ok
+
+    # -------------------------------------------------------- online SGD
+
+    def sgd_update(self, user_id: str, item_id: str, rating: float,
+                   learning_rate: float = 0.1, user_reg: float = 0.0,
+                   item_reg: float = 0.0,
+                   user_mean: Optional[str] = None,
+                   item_mean: Optional[str] = None) -> List[str]:
+        """One online-SGD step, v1 semantics (SGD.java:160-217): query both
+        factor vectors (MEAN-U/MEAN-I fallback for cold starts), compute both
+        updates from the OLD copies, emit + re-ingest the updated rows.
+        Bias is computed but never persisted (SGD.java:209,232)."""
+        u = self.get_vector(als_state_key(user_id, "U"))
+        if u is None:
+            mean = self.get_vector("MEAN-U")
+            u = mean if mean is not None else (
+                [float(x) for x in user_mean.split(";")] if user_mean else None)
+        v = self.get_vector(als_state_key(item_id, "I"))
+        if v is None:
+            mean = self.get_vector("MEAN-I")
+            v = mean if mean is not None else (
+                [float(x) for x in item_mean.split(";")] if item_mean else None)
+        if u is None or v is None:
+            raise KeyError("no factors and no MEAN fallback for "
+                           f"({user_id},{item_id})")
+        err = rating - sum(a * b for a, b in zip(u, v))
+        new_u = [a + learning_rate * (err * b - user_reg * a)
+                 for a, b in zip(u, v)]
+        new_v = [b + learning_rate * (err * a - item_reg * b)
+                 for a, b in zip(u, v)]
+        rows = [
+            f"{user_id},U," + ";".join(java_double_to_string(x) for x in new_u),
+            f"{item_id},I," + ";".join(java_double_to_string(x) for x in new_v),
+        ]
+        # the reference routes these through Kafka back into the consumer;
+        # here the loop closes in-process (same last-writer-wins contract)
+        self.ingest(rows)
+        return rows
+
+    # ------------------------------------------------------- bookkeeping
+
+    def keys(self) -> List[str]:
+        with self._lock:
+            return list(self._payload.keys())
+
+    def snapshot_rows(self) -> List[str]:
+        """All state as model-format text rows (the checkpoint format)."""
+        with self._lock:
+            out = []
+            for key, payload in self._payload.items():
+                entity_id, kind = key.rsplit("-", 1)
+                out.append(f"{entity_id},{kind},{payload}")
+            return out
+
+    def __len__(self) -> int:
+        return len(self._payload)
+
+
+class SVMModelStore:
+    """Keyed store: flat ``"<featureIdx>" -> weight`` or range-partitioned
+    ``"<bucket>" -> "i:w;i:w;..."`` (SVMKafkaConsumer.java:74-92)."""
+
+    def __init__(self):
+        self._payload: Dict[str, str] = {}
+        self._flat: Dict[str, float] = {}                 # id -> weight
+        self._buckets: Dict[str, Dict[str, float]] = {}   # bucket -> id -> w
+        self._lock = threading.RLock()
+
+    def ingest_row(self, row: str) -> str:
+        row = row.strip()
+        key, payload = row.split(",", 1)
+        with self._lock:
+            self._payload[key] = payload
+            if ":" in payload:  # range-partitioned row
+                pairs: Dict[str, float] = {}
+                for item in payload.split(";"):
+                    i, w = item.split(":")
+                    pairs[i] = float(w)
+                self._buckets[key] = pairs
+            else:
+                self._flat[key] = float(payload)
+        return key
+
+    def ingest(self, rows: Iterable[str]) -> int:
+        n = 0
+        for row in rows:
+            if row.strip():
+                self.ingest_row(row)
+                n += 1
+        return n
+
+    def query(self, key: str) -> Optional[Tuple[str, str]]:
+        with self._lock:
+            payload = self._payload.get(key)
+        return None if payload is None else (key, payload)
+
+    def predict(self, pairs: List[Tuple[str, float]],
+                output_decision_function: bool = False,
+                threshold: float = 0.0,
+                range_size: Optional[int] = None
+                ) -> Tuple[float, float, List[str]]:
+        """Sparse-vector scoring.  Flat mode: one lookup per feature
+        (SVMPredict.java:63-86); range mode: one lookup per bucket with
+        bucket = featureId / range (RangePartitionSVMPredict.java:63-101).
+        Returns (prediction, raw value, messages for missing lookups)."""
+        raw = 0.0
+        messages: List[str] = []
+        with self._lock:
+            if range_size is None:
+                for fid, val in pairs:
+                    w = self._flat.get(fid)
+                    if w is None:
+                        messages.append(
+                            f"Could not find the value for feature ID: {fid} ")
+                    else:
+                        raw += w * val
+            else:
+                by_bucket: Dict[str, List[Tuple[str, float]]] = {}
+                for fid, val in pairs:
+                    bucket = str(int(fid) // range_size)
+                    by_bucket.setdefault(bucket, []).append((fid, val))
+                for bucket, feats in by_bucket.items():
+                    bvals = self._buckets.get(bucket)
+                    if bvals is None:
+                        messages.append(
+                            f"could not find model value for key: {bucket}")
+                        continue
+                    for fid, val in feats:
+                        w = bvals.get(fid)
+                        if w is None:
+                            messages.append(
+                                f"feature ID: {fid} not found in bucket: "
+                                f"{bucket}")
+                        else:
+                            raw += w * val
+        pred = raw if output_decision_function else (
+            1.0 if raw > threshold else -1.0)
+        return pred, raw, messages
+
+    def snapshot_rows(self) -> List[str]:
+        with self._lock:
+            return [f"{k},{p}" for k, p in self._payload.items()]
+
+    def keys(self) -> List[str]:
+        with self._lock:
+            return list(self._payload.keys())
+
+    def __len__(self) -> int:
+        return len(self._payload)
+
+
+MEAN_USER_KEY = als_state_key(MEAN_ID, "U")
+MEAN_ITEM_KEY = als_state_key(MEAN_ID, "I")

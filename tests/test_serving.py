@@ -1,0 +1,173 @@
+"""Serving-layer tests: store semantics, REST surface, payload parity,
+online SGD loop, load generators, checkpointing."""
+
+import io
+
+import pytest
+import torch
+from fastapi.testclient import TestClient
+
+from flink_ms_amd.models.als import ALSConfig, train_als
+from flink_ms_amd.data.ratings import RatingsShape, synthetic_ratings
+from flink_ms_amd.serving import ALSModelStore, SVMModelStore, create_app
+from flink_ms_amd.serving.loadgen import (
+    als_predict_random,
+    range_partition_svm_predict,
+    svm_predict_random,
+)
+
+
+@pytest.fixture
+def als_store():
+    store = ALSModelStore(device=torch.device("cpu"))
+    store.ingest([
+        "1,U,0.5;1.0;-0.25",
+        "2,U,1.0;2.0;3.0",
+        "10,I,2.0;0.5;4.0",
+        "MEAN,U,0.75;1.5;1.375",
+        "MEAN,I,2.0;0.5;4.0",
+    ])
+    return store
+
+
+@pytest.fixture
+def svm_store():
+    store = SVMModelStore()
+    store.ingest(["1,0.5", "2,-1.5", "3,2.0"])         # flat rows
+    store.ingest(["0,1:0.5;2:-1.5;3:2.0", "1,1000:0.25;1001:0"])  # range rows
+    return store
+
+
+def test_store_payload_parity(als_store):
+    # payload string returned verbatim, Tuple2(key, payload) shape
+    assert als_store.query("1-U") == ("1-U", "0.5;1.0;-0.25")
+    assert als_store.query("MEAN-I") == ("MEAN-I", "2.0;0.5;4.0")
+    assert als_store.query("99-U") is None  # Optional.empty
+
+
+def test_store_predict_dot(als_store):
+    # dot(U[1], V[10]) = 0.5*2 + 1*0.5 + (-0.25)*4 = 0.5
+    assert als_store.predict("1", "10") == pytest.approx(0.5)
+    assert als_store.predict("99", "10") is None
+
+
+def test_store_hot_swap(als_store):
+    als_store.ingest_row("1,U,1.0;0.0;0.0")  # re-publish == overwrite
+    assert als_store.query("1-U") == ("1-U", "1.0;0.0;0.0")
+    assert als_store.predict("1", "10") == pytest.approx(2.0)
+
+
+def test_store_sgd_update_v1_semantics(als_store):
+    rows = als_store.sgd_update("1", "10", rating=3.0, learning_rate=0.1)
+    # err = 3 - 0.5 = 2.5 ; new_u = u + 0.1*2.5*v ; new_v = v + 0.1*2.5*u(OLD)
+    u = [0.5, 1.0, -0.25]
+    v = [2.0, 0.5, 4.0]
+    exp_u = [a + 0.25 * b for a, b in zip(u, v)]
+    exp_v = [b + 0.25 * a for a, b in zip(u, v)]
+    got_u = [float(x) for x in rows[0].split(",")[2].split(";")]
+    got_v = [float(x) for x in rows[1].split(",")[2].split(";")]
+    assert got_u == pytest.approx(exp_u)
+    assert got_v == pytest.approx(exp_v)
+    # updates persisted (the Kafka round trip collapsed)
+    assert als_store.predict("1", "10") != pytest.approx(0.5)
+
+
+def test_store_sgd_mean_fallback(als_store):
+    rows = als_store.sgd_update("777", "10", rating=4.0)  # unknown user
+    assert rows[0].startswith("777,U,")
+    assert als_store.query("777-U") is not None
+
+
+def test_svm_store_predicts(svm_store):
+    # flat: 1*0.5 + 2*(-1.5) = raw
+    pred, raw, msgs = svm_store.predict([("1", 1.0), ("2", 2.0)],
+                                        output_decision_function=True)
+    assert raw == pytest.approx(0.5 - 3.0)
+    assert pred == pytest.approx(raw) and not msgs
+    # thresholding
+    pred, _, _ = svm_store.predict([("1", 1.0)], threshold=0.0)
+    assert pred == 1.0
+    # missing feature message (SVMPredict.java:76-78)
+    _, _, msgs = svm_store.predict([("42", 1.0)])
+    assert "42" in msgs[0]
+    # range-partitioned path: bucket = id // range
+    pred, raw, msgs = svm_store.predict([("1", 2.0), ("1000", 4.0)],
+                                        output_decision_function=True,
+                                        range_size=1000)
+    assert raw == pytest.approx(2.0 * 0.5 + 4.0 * 0.25) and not msgs
+
+
+def test_rest_surface(tmp_path, als_store, svm_store):
+    app = create_app(als_store, svm_store,
+                     checkpoint_data_uri=str(tmp_path / "ckpt"),
+                     checkpoint_interval_ms=0)
+    c = TestClient(app)
+    assert c.get("/healthz").json()["ok"]
+    # state lookups (queryable-state parity)
+    r = c.get("/state/ALS_MODEL/1-U")
+    assert r.json() == {"key": "1-U", "value": ["1-U", "0.5;1.0;-0.25"]}
+    assert c.get("/state/ALS_MODEL/404-U").status_code == 404
+    assert c.get("/state/SVM_MODEL/2").json()["value"] == ["2", "-1.5"]
+    # predict
+    r = c.get("/als/predict", params={"user": "1", "item": "10"}).json()
+    assert r["found"] and r["prediction"] == pytest.approx(0.5)
+    assert r["formatted"].startswith("ALS Prediction =  0.5")
+    r = c.get("/als/predict", params={"user": "404", "item": "10"}).json()
+    assert not r["found"] and "do not exist" in r["message"]
+    # svm predict
+    r = c.post("/svm/predict", json={"vector": "1:1.0 2:2.0",
+                                     "output_decision_function": True}).json()
+    assert r["raw"] == pytest.approx(-2.5)
+    # ingest
+    r = c.post("/model/als/rows", json={"rows": ["5,U,1.0;1.0;1.0"]})
+    assert r.json()["ingested"] == 1
+    assert c.get("/state/ALS_MODEL/5-U").status_code == 200
+    # sgd
+    r = c.post("/sgd/update", json={"ratings": ["1\t10\t5.0"]}).json()
+    assert r["updated"] == 1 and len(r["rows"]) == 2
+    # mse
+    r = c.post("/mse", json={"ratings": ["1\t10\t1.0", "404\t10\t1.0"]}).json()
+    assert r["scored"] == 1 and r["skipped"] == 1
+    # checkpoint -> restore round trip
+    r = c.post("/checkpoint").json()
+    assert r["written"] == len(als_store) + len(svm_store)
+    import glob
+    files = glob.glob(str(tmp_path / "ckpt" / "als-*.model"))
+    restored = ALSModelStore(device=torch.device("cpu"))
+    with open(files[0]) as f:
+        restored.ingest(f.read().splitlines())
+    assert restored.query("1-U") == als_store.query("1-U")
+
+
+def test_train_to_serve_pipeline(tmp_path):
+    """End-to-end: train ALS -> write model files -> ingest -> serve -> the
+    served predictions equal the trained model's (SURVEY.md data-flow)."""
+    shape = RatingsShape(50, 30, 800)
+    u, i, r = synthetic_ratings(shape, seed=4)
+    model, _ = train_als(u, i, r, 50, 30,
+                         ALSConfig(iterations=3, num_factors=8,
+                                   lambda_=0.1, dtype=torch.float32))
+    uf, itf = io.StringIO(), io.StringIO()
+    model.write(uf, itf)
+    store = ALSModelStore(device=torch.device("cpu"))
+    store.ingest(uf.getvalue().splitlines())
+    store.ingest(itf.getvalue().splitlines())
+    pred = store.predict("0", "0")
+    expected = float(model.user_factors[0].double()
+                     @ model.item_factors[0].double())
+    assert pred == pytest.approx(expected, rel=1e-9)
+
+
+def test_loadgens(als_store, svm_store):
+    res = als_predict_random(num_queries=50, lower_user_id=1, upper_user_id=2,
+                             lower_item_id=10, upper_item_id=10,
+                             store=als_store, seed=1)
+    assert res.misses == 0 and len(res.csv_rows) == 50
+    assert res.csv_rows[0].count(",") == 3  # uId,iId,prediction,millis
+    assert res.summary()["p50_ms"] is not None
+    res = svm_predict_random(max_no_of_features=3, num_queries=20,
+                             store=svm_store, seed=2)
+    assert len(res.csv_rows) == 20
+    res = range_partition_svm_predict(max_no_of_features=3, num_queries=20,
+                                      range_size=1000, store=svm_store, seed=3)
+    assert len(res.csv_rows) == 20 and res.misses == 0
