@@ -1,0 +1,32 @@
+"""Mean-vector job (reference flink-als/.../ALSMeanVector.scala rebuild).
+
+Flags: --type item|user (required), --input, --output.
+"""
+import sys
+
+from ..models.mean_vector import mean_vector_rows
+from ..utils.params import Params
+
+
+def main(argv=None) -> int:
+    params = Params.from_args(sys.argv[1:] if argv is None else argv)
+    t = params.get_required("type")
+    if t == "item":
+        factor_type = "I"
+    elif t == "user":
+        factor_type = "U"
+    else:
+        raise ValueError("specify type as either 'item' or 'user'.")
+    with open(params.get_required("input")) as f:
+        row = mean_vector_rows(f, factor_type)
+    if params.has("output"):
+        with open(params.get("output"), "w") as f:
+            f.write(row + "\n")
+    else:
+        print("Printing results to stdout. Use --output to specify output location")
+        print(row)
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

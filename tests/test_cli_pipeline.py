@@ -1,0 +1,167 @@
+"""End-to-end CLI pipeline test over a LIVE HTTP server: the reference's
+full data-flow topology: train -> mean vector -> publish ->
+serve -> predict / online SGD / MSE / load generators."""
+
+import socket
+import threading
+import time
+
+import pytest
+import torch
+import uvicorn
+
+from flink_ms_amd.cli import (
+    als_mean_vector,
+    als_model_generator,
+    als_predict_random,
+    als_train,
+    mse as mse_cli,
+    producer,
+    serve,
+    sgd as sgd_cli,
+    svm_model_generator,
+    svm_train,
+)
+from flink_ms_amd.data.libsvm import LibSVMShape, synthetic_libsvm, write_libsvm
+from flink_ms_amd.data.ratings import RatingsShape, synthetic_ratings
+from flink_ms_amd.serving.client import QueryClientHelper
+from flink_ms_amd.utils.params import Params
+
+
+@pytest.fixture(scope="module")
+def server():
+    port = _free_port()
+    app = serve.build_app(Params({}))
+    config = uvicorn.Config(app, host="127.0.0.1", port=port,
+                            log_level="error")
+    srv = uvicorn.Server(config)
+    th = threading.Thread(target=srv.run, daemon=True)
+    th.start()
+    for _ in range(100):
+        try:
+            with QueryClientHelper("127.0.0.1", port, 2) as c:
+                c._client.get(c.base + "/healthz").raise_for_status()
+            break
+        except Exception:
+            time.sleep(0.1)
+    yield port
+    srv.should_exit = True
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+@pytest.mark.timeout(300)
+def test_full_pipeline(tmp_path, server, capsys):
+    port = str(server)
+    # ---- 1. training data + ALS training job
+    shape = RatingsShape(60, 40, 1200)
+    u, i, r = synthetic_ratings(shape, seed=13)
+    ratings_csv = tmp_path / "ratings.csv"
+    with open(ratings_csv, "w") as f:
+        f.write("userId,movieId,rating\n")
+        for a, b, c in zip(u.tolist(), i.tolist(), r.tolist()):
+            f.write(f"{a},{b},{c}\n")
+    ufile, ifile = tmp_path / "userFactors", tmp_path / "itemFactors"
+    assert als_train.main([
+        "--input", str(ratings_csv), "--iterations", "3",
+        "--numFactors", "8", "--lambda", "0.1",
+        "--userFactors", str(ufile), "--itemFactors", str(ifile)]) == 0
+
+    # ---- 2. mean-vector job (cold-start rows)
+    umean, imean = tmp_path / "umean", tmp_path / "imean"
+    assert als_mean_vector.main(["--type", "user", "--input", str(ufile),
+                                 "--output", str(umean)]) == 0
+    assert als_mean_vector.main(["--type", "item", "--input", str(ifile),
+                                 "--output", str(imean)]) == 0
+    assert open(umean).read().startswith("MEAN,U,")
+
+    # ---- 3. publish to the serving job (producer == Kafka path)
+    for path in (ufile, ifile, umean, imean):
+        assert producer.main(["--input", str(path), "--model", "als",
+                              "--server", "127.0.0.1", "--port", port]) == 0
+
+    # ---- 4. point queries return trained factors byte-identically
+    with QueryClientHelper("127.0.0.1", int(port)) as client:
+        hit = client.query_state("ALS_MODEL", "0-U")
+        assert hit is not None
+        assert hit[1] == open(ufile).read().splitlines()[0].split(",", 2)[2]
+        pred = client.als_predict("0", "0")
+        assert pred["found"]
+
+    # ---- 5. load generator against the live server
+    assert als_predict_random.main([
+        "--jobManagerHost", "127.0.0.1", "--jobManagerPort", port,
+        "--numQueries", "30", "--lowerUserId", "0", "--upperUserId", "59",
+        "--lowerItemId", "0", "--upperItemId", "39",
+        "--outputFile", str(tmp_path / "lat.csv")]) == 0
+    lat = open(tmp_path / "lat.csv").read().splitlines()
+    assert lat[0] == "uId,iId,prediction,millis" and len(lat) > 1
+
+    # ---- 6. online SGD job (once mode), closing the loop via the server
+    sgd_file = tmp_path / "stream.tsv"
+    with open(sgd_file, "w") as f:
+        f.write("0\t0\t4.5\n999\t0\t3.0\n")  # known pair + cold-start user
+    before = QueryClientHelper("127.0.0.1", int(port)).query_state(
+        "ALS_MODEL", "0-U")
+    assert sgd_cli.main(["--input", str(sgd_file), "--mode", "once",
+                         "--jobManagerHost", "127.0.0.1",
+                         "--jobManagerPort", port]) == 0
+    with QueryClientHelper("127.0.0.1", int(port)) as client:
+        after = client.query_state("ALS_MODEL", "0-U")
+        assert after is not None and after[1] != before[1]
+        assert client.query_state("ALS_MODEL", "999-U") is not None  # MEAN init
+
+    # ---- 7. MSE job
+    mse_file = tmp_path / "test.tsv"
+    with open(mse_file, "w") as f:
+        for a, b, c in zip(u.tolist()[:100], i.tolist()[:100], r.tolist()[:100]):
+            f.write(f"{a}\t{b}\t{c}\n")
+    assert mse_cli.main(["--input", str(mse_file),
+                         "--jobManagerHost", "127.0.0.1",
+                         "--jobManagerPort", port]) == 0
+    assert "MSE = " in capsys.readouterr().out
+
+
+@pytest.mark.timeout(300)
+def test_svm_pipeline(tmp_path, server):
+    port = str(server)
+    csr, y = synthetic_libsvm(LibSVMShape(120, 30, 6), seed=9, separable=True)
+    train_file = tmp_path / "train.libsvm"
+    write_libsvm(str(train_file), csr, y)
+    flat_model = tmp_path / "svm_flat.model"
+    assert svm_train.main(["--training", str(train_file), "--iteration", "3",
+                           "--output", str(flat_model)]) == 0
+    rows = open(flat_model).read().splitlines()
+    assert len(rows) == 30 and rows[0].startswith("1,")
+    part_model = tmp_path / "svm_part.model"
+    assert svm_train.main(["--training", str(train_file), "--iteration", "3",
+                           "--partition", "--range", "10",
+                           "--output", str(part_model)]) == 0
+    assert open(part_model).read().startswith("0,")
+    assert producer.main(["--input", str(flat_model), "--model", "svm",
+                          "--server", "127.0.0.1", "--port", port]) == 0
+    with QueryClientHelper("127.0.0.1", int(port)) as client:
+        assert client.query_state("SVM_MODEL", "1") is not None
+        resp = client.svm_predict("1:1.0 2:0.5",
+                                  output_decision_function=True)
+        w1 = float(rows[0].split(",")[1])
+        w2 = float(rows[1].split(",")[1])
+        assert resp["raw"] == pytest.approx(w1 + 0.5 * w2, rel=1e-9)
+
+
+def test_generator_clis(tmp_path):
+    out = tmp_path / "als.model"
+    assert als_model_generator.main(["--numUsers", "5", "--numItems", "3",
+                                     "--latentFactors", "4",
+                                     "--output", str(out)]) == 0
+    assert len(open(out).read().splitlines()) == 8
+    out2 = tmp_path / "svm.model"
+    assert svm_model_generator.main(["--numFeatures", "50", "--range", "10",
+                                     "--output", str(out2)]) == 0
+    assert len(open(out2).read().splitlines()) == 5

@@ -34,7 +34,13 @@ def als_store():
 def svm_store():
     store = SVMModelStore()
     store.ingest(["1,0.5", "2,-1.5", "3,2.0"])         # flat rows
-    store.ingest(["0,1:0.5;2:-1.5;3:2.0", "1,1000:0.25;1001:0"])  # range rows
+    return store
+
+
+@pytest.fixture
+def svm_range_store():
+    store = SVMModelStore()
+    store.ingest(["0,1:0.5;2:-1.5;3:2.0", "1,1000:0.25;1001:0"])
     return store
 
 
@@ -78,7 +84,7 @@ def test_store_sgd_mean_fallback(als_store):
     assert als_store.query("777-U") is not None
 
 
-def test_svm_store_predicts(svm_store):
+def test_svm_store_predicts(svm_store, svm_range_store):
     # flat: 1*0.5 + 2*(-1.5) = raw
     pred, raw, msgs = svm_store.predict([("1", 1.0), ("2", 2.0)],
                                         output_decision_function=True)
@@ -91,7 +97,7 @@ def test_svm_store_predicts(svm_store):
     _, _, msgs = svm_store.predict([("42", 1.0)])
     assert "42" in msgs[0]
     # range-partitioned path: bucket = id // range
-    pred, raw, msgs = svm_store.predict([("1", 2.0), ("1000", 4.0)],
+    pred, raw, msgs = svm_range_store.predict([("1", 2.0), ("1000", 4.0)],
                                         output_decision_function=True,
                                         range_size=1000)
     assert raw == pytest.approx(2.0 * 0.5 + 4.0 * 0.25) and not msgs
@@ -158,7 +164,7 @@ def test_train_to_serve_pipeline(tmp_path):
     assert pred == pytest.approx(expected, rel=1e-9)
 
 
-def test_loadgens(als_store, svm_store):
+def test_loadgens(als_store, svm_store, svm_range_store):
     res = als_predict_random(num_queries=50, lower_user_id=1, upper_user_id=2,
                              lower_item_id=10, upper_item_id=10,
                              store=als_store, seed=1)
@@ -169,5 +175,6 @@ def test_loadgens(als_store, svm_store):
                              store=svm_store, seed=2)
     assert len(res.csv_rows) == 20
     res = range_partition_svm_predict(max_no_of_features=3, num_queries=20,
-                                      range_size=1000, store=svm_store, seed=3)
+                                      range_size=1000, store=svm_range_store,
+                                      seed=3)
     assert len(res.csv_rows) == 20 and res.misses == 0
