@@ -1,0 +1,87 @@
+#!/usr/bin/env python3
+"""Serving-latency benchmark (BASELINE config 5): p50/p95 point-query and
+prediction latency of the queryable-state surface, in-process and over live
+HTTP, on a synthetic model (reference harness: ALSPredictRandom /
+RangePartitionSVMPredict latency CSVs)."""
+
+import json
+import os
+import socket
+import sys
+import threading
+import time
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+
+from flink_ms_amd.serving.app import create_app
+from flink_ms_amd.serving.client import QueryClientHelper
+from flink_ms_amd.serving.loadgen import (
+    als_predict_random,
+    range_partition_svm_predict,
+)
+from flink_ms_amd.serving.store import ALSModelStore, SVMModelStore
+from flink_ms_amd.utils.textio import als_factor_row, svm_range_row
+
+
+def build_stores(num_users=20000, num_items=5000, k=64, num_features=47236,
+                 range_size=1000):
+    import numpy as np
+    rng = np.random.default_rng(42)
+    als = ALSModelStore()
+    for uid in range(num_users):
+        als.ingest_row(als_factor_row(uid, "U", rng.random(k) * 0.5))
+    for iid in range(num_items):
+        als.ingest_row(als_factor_row(iid, "I", rng.random(k) * 0.5))
+    svm = SVMModelStore()
+    for bucket in range((num_features + range_size - 1) // range_size):
+        start = bucket * range_size
+        w = rng.random(range_size) * 2 - 1
+        svm.ingest_row(svm_range_row(
+            bucket, [(start + j, w[j]) for j in range(range_size)]))
+    return als, svm
+
+
+def main():
+    t0 = time.perf_counter()
+    als, svm = build_stores()
+    print(f"store build: {time.perf_counter()-t0:.1f}s "
+          f"({len(als)} ALS keys, {len(svm)} SVM buckets)", flush=True)
+
+    results = {}
+    # in-process (store-direct) latency
+    r = als_predict_random(num_queries=5000, upper_user_id=19999,
+                           upper_item_id=4999, store=als, seed=1)
+    results["als_inproc"] = r.summary()
+    r = range_partition_svm_predict(max_no_of_features=47236, num_queries=200,
+                                    range_size=1000, store=svm, seed=2,
+                                    min_percentage_of_features=0)
+    results["svm_range_inproc"] = r.summary()
+
+    # live-HTTP latency (uvicorn + httpx: the full REST path)
+    import uvicorn
+    s = socket.socket(); s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]; s.close()
+    app = create_app(als, svm)
+    srv = uvicorn.Server(uvicorn.Config(app, host="127.0.0.1", port=port,
+                                        log_level="error"))
+    threading.Thread(target=srv.run, daemon=True).start()
+    client = QueryClientHelper("127.0.0.1", port)
+    for _ in range(100):
+        try:
+            client._client.get(client.base + "/healthz").raise_for_status()
+            break
+        except Exception:
+            time.sleep(0.1)
+    r = als_predict_random(num_queries=2000, upper_user_id=19999,
+                           upper_item_id=4999, client=client, seed=3)
+    results["als_http"] = r.summary()
+    r = range_partition_svm_predict(max_no_of_features=47236, num_queries=50,
+                                    range_size=1000, client=client, seed=4,
+                                    min_percentage_of_features=0)
+    results["svm_range_http"] = r.summary()
+    srv.should_exit = True
+    print(json.dumps(results, indent=1), flush=True)
+
+
+if __name__ == "__main__":
+    main()
