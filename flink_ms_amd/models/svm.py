@@ -93,6 +93,10 @@ class SVMTrainer:
                                  device=dev)
         self.norms_sq = ops.csr_row_norms_sq(self.csr)
         self._gen = torch.Generator(device="cpu").manual_seed(self.cfg.seed)
+        # one fixed shuffled visit order: the local solver is hogwild-async,
+        # so per-pass reshuffles buy nothing but a host randperm + H2D copy
+        self._perm = torch.randperm(
+            local_csr.num_rows, generator=self._gen).to(torch.int32).to(dev)
 
     def step(self) -> float:
         """One outer CoCoA iteration; returns wall seconds (max over ranks)."""
@@ -102,11 +106,9 @@ class SVMTrainer:
         v = self.w.clone()
         a0 = self.alpha.clone()
         for _ in range(cfg.local_iterations):
-            perm = torch.randperm(self.csr.num_rows, generator=self._gen
-                                  ).to(torch.int32).to(ctx.device)
             ops.sdca_pass(self.csr, self.y, self.alpha, v,
                           cfg.regularization, self.n_global,
-                          norms_sq=self.norms_sq, perm=perm)
+                          norms_sq=self.norms_sq, perm=self._perm)
         K = max(ctx.world_size, 1)
         dw = v - self.w
         ctx.all_reduce_(dw)
