@@ -33,6 +33,7 @@ def run_once(client: QueryClientHelper, rows, params, out_file=None) -> int:
         item_regularization=params.get_float("itemRegularization", 0.0),
         user_mean=params.get("userMean"),
         item_mean=params.get("itemMean"),
+        v0_semantics=params.get_bool("v0", False),  # SGDV0.java variant
     )
     if not rows:
         return 0

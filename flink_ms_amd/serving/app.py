@@ -59,6 +59,7 @@ class SGDBody(BaseModel):
     item_regularization: float = 0.0
     user_mean: Optional[str] = None
     item_mean: Optional[str] = None
+    v0_semantics: bool = False  # SGDV0.java in-place updates + NaN filter
 
 
 class MSEBody(BaseModel):
@@ -205,7 +206,8 @@ def create_app(als_store: Optional[ALSModelStore] = None,
                 rows = als.sgd_update(
                     u, i, float(r), body.learning_rate,
                     body.user_regularization, body.item_regularization,
-                    body.user_mean, body.item_mean)
+                    body.user_mean, body.item_mean,
+                    v0_semantics=body.v0_semantics)
             except KeyError as e:
                 raise HTTPException(400, str(e))
             for out_row in rows:

@@ -149,11 +149,18 @@ class ALSModelStore:
                    learning_rate: float = 0.1, user_reg: float = 0.0,
                    item_reg: float = 0.0,
                    user_mean: Optional[str] = None,
-                   item_mean: Optional[str] = None) -> List[str]:
-        """One online-SGD step, v1 semantics (SGD.java:160-217): query both
-        factor vectors (MEAN-U/MEAN-I fallback for cold starts), compute both
-        updates from the OLD copies, emit + re-ingest the updated rows.
-        Bias is computed but never persisted (SGD.java:209,232)."""
+                   item_mean: Optional[str] = None,
+                   v0_semantics: bool = False) -> List[str]:
+        """One online-SGD step: query both factor vectors (MEAN-U/MEAN-I
+        fallback for cold starts), update, emit + re-ingest the rows.
+
+        Default is v1 "simultaneous" semantics — both updates computed from
+        the OLD copies (SGD.java:199-207).  ``v0_semantics`` reproduces
+        SGDV0.java:188-197 instead: in-place updates, so the item update
+        sees the already-updated user vector, and NaN-containing output
+        rows are dropped before ingest (SGDV0.java:207-226).  Bias is
+        computed but never persisted in either version (SGD.java:209,232).
+        """
         u = self.get_vector(als_state_key(user_id, "U"))
         if u is None:
             mean = self.get_vector("MEAN-U")
@@ -168,14 +175,23 @@ class ALSModelStore:
             raise KeyError("no factors and no MEAN fallback for "
                            f"({user_id},{item_id})")
         err = rating - sum(a * b for a, b in zip(u, v))
-        new_u = [a + learning_rate * (err * b - user_reg * a)
-                 for a, b in zip(u, v)]
-        new_v = [b + learning_rate * (err * a - item_reg * b)
-                 for a, b in zip(u, v)]
+        if v0_semantics:
+            # in-place: q-update reads the NEW p (SGDV0 quirk)
+            new_u = [a + learning_rate * (err * b - user_reg * a)
+                     for a, b in zip(u, v)]
+            new_v = [b + learning_rate * (err * a - item_reg * b)
+                     for a, b in zip(new_u, v)]
+        else:
+            new_u = [a + learning_rate * (err * b - user_reg * a)
+                     for a, b in zip(u, v)]
+            new_v = [b + learning_rate * (err * a - item_reg * b)
+                     for a, b in zip(u, v)]
         rows = [
             f"{user_id},U," + ";".join(java_double_to_string(x) for x in new_u),
             f"{item_id},I," + ";".join(java_double_to_string(x) for x in new_v),
         ]
+        if v0_semantics:  # NaN output filter (SGDV0.java:207-226)
+            rows = [r for r in rows if "NaN" not in r]
         # the reference routes these through Kafka back into the consumer;
         # here the loop closes in-process (same last-writer-wins contract)
         self.ingest(rows)

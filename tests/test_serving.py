@@ -178,3 +178,26 @@ def test_loadgens(als_store, svm_store, svm_range_store):
                                       range_size=1000, store=svm_range_store,
                                       seed=3)
     assert len(res.csv_rows) == 20 and res.misses == 0
+
+
+def test_store_sgd_v0_semantics(als_store):
+    """SGDV0: in-place update — the item update sees the NEW user vector
+    (SGDV0.java:188-197)."""
+    u = [0.5, 1.0, -0.25]
+    v = [2.0, 0.5, 4.0]
+    err = 3.0 - 0.5
+    new_u = [a + 0.1 * (err * b) for a, b in zip(u, v)]
+    exp_v = [b + 0.1 * (err * a) for a, b in zip(new_u, v)]
+    rows = als_store.sgd_update("1", "10", rating=3.0, learning_rate=0.1,
+                                v0_semantics=True)
+    got_v = [float(x) for x in rows[1].split(",")[2].split(";")]
+    assert got_v == pytest.approx(exp_v)
+
+
+def test_store_sgd_v0_nan_filter():
+    store = ALSModelStore(device=torch.device("cpu"))
+    store.ingest(["1,U,Infinity", "10,I,Infinity"])  # engineered NaN source
+    rows = store.sgd_update("1", "10", rating=1.0, learning_rate=0.1,
+                            v0_semantics=True)
+    assert rows == []  # NaN rows dropped, state unchanged
+    assert store.query("1-U") == ("1-U", "Infinity")
