@@ -89,7 +89,7 @@ def als_solve_side(
         ob = out_bf16 if out_bf16 is not None else _empty(fac.device)
         ro = row_order if row_order is not None else _empty(fac.device)
         korig = other_factors.shape[1]
-        if fused or k > 64:
+        if fused:
             out = torch.empty(csr.num_rows, k, dtype=torch.float32,
                               device=fac.device)
             ops.als_solve_fused(csr.indptr, csr.indices, csr.values, fac,
@@ -102,7 +102,12 @@ def als_solve_side(
             ops.gramian(csr.indptr, csr.indices, csr.values, fac, A, b,
                         float(reg), _stream())
             out = torch.empty_like(b)
-            ops.ldl_solve_wave(A, b, out, ob, _stream())
+            if k <= 64:
+                ops.ldl_solve_wave(A, b, out, ob, _stream())
+            else:
+                ops.cholesky_solve(A, b, out, _stream())
+                if out_bf16 is not None:
+                    out_bf16.copy_(out.to(torch.bfloat16))
         return out[:, :korig] if korig != k else out
     return reference.als_solve_side_reference(csr, other_factors, reg)
 

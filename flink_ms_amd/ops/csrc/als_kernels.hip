@@ -256,46 +256,79 @@ __global__ void k_ldl_solve_wave(const float* __restrict__ A_in,
     for (int i = lane; i < K * K; i += 64)
         A[(i / K) * LDA + (i % K)] = src[i];
     float x0 = (lane < K) ? b_in[e * K + lane] : 0.0f;
-    // In-place LDL^T elimination, rank-4 panels; column values stay raw.
-    // Lanes run in wave lockstep, so panel-internal cross-lane dependencies
-    // (lane r reads lane c's row element) are ordered by program order.
-    // Cells (r,c) with c > r are never-read upper-triangle scratch, so the
-    // update runs guard-free over the full trailing range (per-element
-    // guards make hipcc branch around every load: cdna_hip_programming.md
-    // §5 ".s-level traps" (c)).  The joint update amortizes one row
-    // read/write over the panel's 4 columns.
-    if (lane < K) {
-        float* rowp = A + lane * LDA;
-        for (int p = 0; p < K; p += 4) {
-            float lscl[4];
-            // factor the 4-column panel (small triangle, lockstep)
+    // In-place LDL^T elimination, 16-column panels with MFMA trailing
+    // updates (cdna_hip_programming.md §6 G10: the trailing update
+    // A22 -= L21 D^-1 L21^T is matmul-shaped, so it runs on the f32 matrix
+    // cores: v_mfma_f32_16x16x4_f32, exact fp32).  Per panel:
+    //   (a) factor the panel slab in place, column by column, each lane
+    //       updating its own row's panel segment (columns stay raw L*D;
+    //       over-length quads CLAMP to the panel edge -- duplicate
+    //       same-value writes, never out-of-segment ones);
+    //   (b) subtract the panel's outer product from the trailing block as
+    //       16x16 MFMA tiles (B operand scaled by -1/D on load).  Tiles on
+    //       or above the diagonal write upper-triangle scratch only.
+    // Lanes run in wave lockstep, so cross-lane panel dependencies are
+    // ordered by program order; no barriers.
+    // NOTE: the MFMA tiles need ALL 64 lanes live (every lane carries one
+    // (i,k) fragment element regardless of K), so only the row-ownership
+    // panel factorization is masked to lane < K; every address in the
+    // MFMA phase is bounded by K for all lanes.
+    {
+        constexpr int NP = K / 16;
+        float* rowp = A + lane * LDA;   // only dereferenced when lane < K
+        const int li = lane & 15, g4 = lane >> 4;
+        for (int pi = 0; pi < NP; ++pi) {
+            const int P0 = 16 * pi;
+            // (a) panel factorization, fully in registers: each lane holds
+            // its row's 16-column panel segment; the pivot column A[c][j]
+            // lives in lane c's seg[jj], broadcast by __shfl -- the inner
+            // rank-1 update is shfl+fma with ZERO LDS traffic.  (For the
+            // last panel j runs to P0+14 = K-2, same bound.)
+            if (lane < K) {
+                float seg[16];
 #pragma unroll
-            for (int jj = 0; jj < 4; ++jj) {
-                const int j = p + jj;
-                const float dj = A[j * LDA + j];
-                const float dinv = dj > 0.0f ? 1.0f / dj : 0.0f;
-                lscl[jj] = rowp[j] * dinv;
+                for (int cc = 0; cc < 16; ++cc) seg[cc] = rowp[P0 + cc];
 #pragma unroll
-                for (int cc = jj + 1; cc < 4; ++cc)
-                    rowp[p + cc] -= lscl[jj] * A[(p + cc) * LDA + j];
-            }
-            // joint trailing update: rows x remaining columns, 4 deep
-            for (int c0 = p + 4; c0 < K; c0 += 4) {
-                float rowv[4], colv[4][4];
+                for (int jj = 0; jj < 15; ++jj) {
+                    const float dj = __shfl(seg[jj], P0 + jj, WAVE);
+                    const float dinv = dj > 0.0f ? 1.0f / dj : 0.0f;
+                    const float lscl = seg[jj] * dinv;
 #pragma unroll
-                for (int m = 0; m < 4; ++m) {
-                    rowv[m] = rowp[c0 + m];
-#pragma unroll
-                    for (int jj = 0; jj < 4; ++jj)
-                        colv[m][jj] = A[(c0 + m) * LDA + p + jj];
+                    for (int cc = jj + 1; cc < 16; ++cc) {
+                        const float colj = __shfl(seg[jj], P0 + cc, WAVE);
+                        seg[cc] -= lscl * colj;
+                    }
                 }
 #pragma unroll
-                for (int m = 0; m < 4; ++m) {
-                    float v = rowv[m];
+                for (int cc = 0; cc < 16; ++cc) rowp[P0 + cc] = seg[cc];
+            }
+            if (pi == NP - 1) break;
+            // (b) MFMA trailing update, whole wave.  -1/D per contraction
+            // column, col(kk) = P0 + 4*kk + g4.
+            float ndk[4];
 #pragma unroll
-                    for (int jj = 0; jj < 4; ++jj)
-                        v -= lscl[jj] * colv[m][jj];
-                    rowp[c0 + m] = v;
+            for (int kk = 0; kk < 4; ++kk) {
+                const int c = P0 + 4 * kk + g4;
+                const float d = A[c * LDA + c];
+                ndk[kk] = d > 0.0f ? -1.0f / d : 0.0f;
+            }
+            for (int rb = pi + 1; rb < NP; ++rb) {
+                for (int cb = pi + 1; cb <= rb; ++cb) {
+                    f32x4 acc;   // C tile: D map row=(l>>4)*4+r, col=l&15
+#pragma unroll
+                    for (int r = 0; r < 4; ++r)
+                        acc[r] = A[(rb * 16 + g4 * 4 + r) * LDA + cb * 16 + li];
+#pragma unroll
+                    for (int kk = 0; kk < 4; ++kk) {
+                        const int pc = P0 + 4 * kk + g4;
+                        const float a = A[(rb * 16 + li) * LDA + pc];
+                        const float b = A[(cb * 16 + li) * LDA + pc] * ndk[kk];
+                        acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc,
+                                                                   0, 0, 0);
+                    }
+#pragma unroll
+                    for (int r = 0; r < 4; ++r)
+                        A[(rb * 16 + g4 * 4 + r) * LDA + cb * 16 + li] = acc[r];
                 }
             }
         }

@@ -201,3 +201,15 @@ def test_store_sgd_v0_nan_filter():
                             v0_semantics=True)
     assert rows == []  # NaN rows dropped, state unchanged
     assert store.query("1-U") == ("1-U", "Infinity")
+
+
+@pytest.mark.gpu
+def test_store_predict_batch_gpu():
+    """The store's device bf16 mirror + K5 kernel batch path."""
+    store = ALSModelStore(device=torch.device("cuda:0"))
+    store.ingest(["1,U,0.5;1.0;-0.25;0.5", "2,U,1.0;2.0;3.0;0.0",
+                  "10,I,2.0;0.5;4.0;1.0"])
+    preds, ok = store.predict_batch(["1", "2", "404"], ["10", "10", "10"])
+    assert ok.tolist() == [True, True, False]
+    assert preds[0].item() == pytest.approx(0.5 + 0.5, rel=1e-2)
+    assert preds[1].item() == pytest.approx(1.0 * 2 + 2 * 0.5 + 3 * 4, rel=1e-2)
