@@ -72,6 +72,7 @@ def als_solve_side(
     out_bf16: Optional[torch.Tensor] = None,
     row_order: Optional[torch.Tensor] = None,
     fused: bool = False,
+    slab_rows: Optional[int] = None,
 ) -> torch.Tensor:
     """One ALS half-iteration: solve every row entity of ``csr`` against the
     opposite side's factors.  Returns fp32 [num_rows, k]; optionally also
@@ -95,19 +96,29 @@ def als_solve_side(
             ops.als_solve_fused(csr.indptr, csr.indices, csr.values, fac,
                                 out, ob, ro, float(reg), _stream())
         else:
-            A = torch.empty(csr.num_rows, k, k, dtype=torch.float32,
+            # slab the normal equations: A is nrows*k*k fp32, which at the
+            # 1B-rating configs would exceed HBM if materialized whole.
+            # A sliced indptr keeps GLOBAL offsets into indices/values, so
+            # each slab call reuses the full nnz arrays untouched.
+            slab = slab_rows or max(1, min(csr.num_rows,
+                                           (1 << 30) // (k * k * 4)))
+            out = torch.empty(csr.num_rows, k, dtype=torch.float32,
+                              device=fac.device)
+            A = torch.empty(slab, k, k, dtype=torch.float32,
                             device=fac.device)
-            b = torch.empty(csr.num_rows, k, dtype=torch.float32,
-                            device=fac.device)
-            ops.gramian(csr.indptr, csr.indices, csr.values, fac, A, b,
-                        float(reg), _stream())
-            out = torch.empty_like(b)
-            if k <= 64:
-                ops.ldl_solve_wave(A, b, out, ob, _stream())
-            else:
-                ops.cholesky_solve(A, b, out, _stream())
-                if out_bf16 is not None:
-                    out_bf16.copy_(out.to(torch.bfloat16))
+            b = torch.empty(slab, k, dtype=torch.float32, device=fac.device)
+            for s in range(0, csr.num_rows, slab):
+                e = min(s + slab, csr.num_rows)
+                As, bs = A[: e - s], b[: e - s]
+                ops.gramian(csr.indptr[s:e + 1], csr.indices, csr.values,
+                            fac, As, bs, float(reg), _stream())
+                obs = ob[s:e] if ob.numel() > 0 else ob
+                if k <= 64:
+                    ops.ldl_solve_wave(As, bs, out[s:e], obs, _stream())
+                else:
+                    ops.cholesky_solve(As, bs, out[s:e], _stream())
+                    if out_bf16 is not None:
+                        out_bf16[s:e].copy_(out[s:e].to(torch.bfloat16))
         return out[:, :korig] if korig != k else out
     return reference.als_solve_side_reference(csr, other_factors, reg)
 

@@ -172,3 +172,15 @@ def test_ldl_wave_solver_parity(gpu, k):
     torch.cuda.synchronize()
     assert torch.allclose(x, x_ref, atol=1e-3, rtol=1e-3)
     assert torch.allclose(xb.to(torch.float32), x, atol=1e-1, rtol=2e-2)
+
+
+def test_slabbed_solve_matches_unslabbed(gpu):
+    """The slab-chunked modular path (sliced indptr, global nnz offsets)
+    must equal the single-slab result."""
+    csr = _rand_csr(rows=500, cols=300, nnz=30_000, seed=33, device=gpu)
+    fac = (torch.randn(300, 64, generator=torch.Generator().manual_seed(5))
+           * 0.5).to(torch.bfloat16).to(gpu)
+    full = ops.als_solve_side(csr, fac, reg=0.4)
+    slabbed = ops.als_solve_side(csr, fac, reg=0.4, slab_rows=77)
+    torch.cuda.synchronize()
+    assert torch.equal(full, slabbed)
