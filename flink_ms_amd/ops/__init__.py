@@ -90,7 +90,8 @@ def als_solve_side(
         ob = out_bf16 if out_bf16 is not None else _empty(fac.device)
         ro = row_order if row_order is not None else _empty(fac.device)
         korig = other_factors.shape[1]
-        if fused:
+        if fused or k > 64:
+            # k>64: fused wins (no nrows*k*k A round-trip through HBM)
             out = torch.empty(csr.num_rows, k, dtype=torch.float32,
                               device=fac.device)
             ops.als_solve_fused(csr.indptr, csr.indices, csr.values, fac,

@@ -58,21 +58,9 @@ __global__ void k_als_solve_fused(const long long* __restrict__ indptr,
 
     float* A = (float*)smem;
     float* b = A + K * (K + 1);
-    cholesky_lds<K>(A);
-
-    const int lane = threadIdx.x & 63;
-    if (threadIdx.x < 64) {
-        float x0, x1;
-        solve_lds<K>(A, b, lane, x0, x1);
-        if (lane < K) {
-            out_f32[row * K + lane] = x0;
-            if (out_bf16) out_bf16[row * K + lane] = f2bf(x0);
-        }
-        if (K > 64 && lane + 64 < K) {
-            out_f32[row * K + lane + 64] = x1;
-            if (out_bf16) out_bf16[row * K + lane + 64] = f2bf(x1);
-        }
-    }
+    cholesky_lds<K>(A, b + 2 * K);
+    solve_lds_block<K>(A, b, b + 2 * K, out_f32 + row * K,
+                       out_bf16 ? out_bf16 + row * K : nullptr);
 }
 
 // Standalone K1 (for parity tests / modular path): writes dense A and b.
@@ -108,7 +96,7 @@ __global__ void k_gramian(const long long* __restrict__ indptr,
 // Standalone K2: batched SPD solve from dense global A/b.
 // phases bitmask (debug/ablation): 1 = eliminate, 2 = solve
 template <int KT>
-__launch_bounds__(256)
+__launch_bounds__(256, 2)
 __global__ void k_cholesky_solve(const float* __restrict__ A_in,  // [n][K][K]
                                  const float* __restrict__ b_in,  // [n][K]
                                  float* __restrict__ x_out,       // [n][K]
@@ -124,12 +112,11 @@ __global__ void k_cholesky_solve(const float* __restrict__ A_in,  // [n][K][K]
         A[(i / K) * (K + 1) + (i % K)] = A_in[row * K * K + i];
     for (int c = tid; c < K; c += 256) b[c] = b_in[row * K + c];
     __syncthreads();
-    if (phases & 1) cholesky_lds<K>(A);
-    if (tid < 64) {
-        float x0 = b[tid & 63], x1 = 0.0f;
-        if (phases & 2) solve_lds<K>(A, b, tid, x0, x1);
-        if (tid < K) x_out[row * K + tid] = x0;
-        if (K > 64 && tid + 64 < K) x_out[row * K + tid + 64] = x1;
+    if (phases & 1) cholesky_lds<K>(A, b + 2 * K);
+    if (phases & 2) {
+        solve_lds_block<K>(A, b, b + 2 * K, x_out + row * K, nullptr);
+    } else if (tid < K) {
+        x_out[row * K + tid] = b[tid];
     }
 }
 
