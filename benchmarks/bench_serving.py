@@ -41,6 +41,57 @@ def build_stores(num_users=20000, num_items=5000, k=64, num_features=47236,
     return als, svm
 
 
+def fixed_qps_als(port, qps, duration_s=5.0, workers=8,
+                  num_users=20000, num_items=5000):
+    """Open-loop fixed-QPS load: queries dispatched on a fixed schedule by a
+    worker pool; reports achieved rate + latency percentiles (the BASELINE
+    'p50 at fixed QPS' serving metric)."""
+    import queue as queue_mod
+    import random
+    import threading as th
+
+    rng = random.Random(7)
+    n = int(qps * duration_s)
+    t0 = time.perf_counter() + 0.2
+    jobs = queue_mod.Queue()
+    for k in range(n):
+        jobs.put((t0 + k / qps, rng.randrange(num_users),
+                  rng.randrange(num_items)))
+    lats = []
+    lock = th.Lock()
+
+    def worker():
+        client = QueryClientHelper("127.0.0.1", port)
+        while True:
+            try:
+                due, u, i = jobs.get_nowait()
+            except queue_mod.Empty:
+                client.close()
+                return
+            now = time.perf_counter()
+            if due > now:
+                time.sleep(due - now)
+            s0 = time.perf_counter()
+            client.als_predict(str(u), str(i))
+            with lock:
+                lats.append((time.perf_counter() - s0) * 1000.0)
+
+    threads = [th.Thread(target=worker) for _ in range(workers)]
+    start = time.perf_counter()
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    wall = time.perf_counter() - start
+    lats.sort()
+
+    def pct(p):
+        return lats[min(len(lats) - 1, int(p * len(lats)))] if lats else None
+    return {"target_qps": qps, "achieved_qps": len(lats) / wall,
+            "queries": len(lats), "p50_ms": pct(0.5), "p95_ms": pct(0.95),
+            "p99_ms": pct(0.99)}
+
+
 def main():
     t0 = time.perf_counter()
     als, svm = build_stores()
@@ -79,6 +130,8 @@ def main():
                                     range_size=1000, client=client, seed=4,
                                     min_percentage_of_features=0)
     results["svm_range_http"] = r.summary()
+    for qps in (500, 2000):
+        results[f"als_http_qps{qps}"] = fixed_qps_als(port, qps)
     srv.should_exit = True
     print(json.dumps(results, indent=1), flush=True)
 

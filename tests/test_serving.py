@@ -213,3 +213,43 @@ def test_store_predict_batch_gpu():
     assert ok.tolist() == [True, True, False]
     assert preds[0].item() == pytest.approx(0.5 + 0.5, rel=1e-2)
     assert preds[1].item() == pytest.approx(1.0 * 2 + 2 * 0.5 + 3 * 4, rel=1e-2)
+
+
+def test_store_concurrent_ingest_and_query():
+    """The serving job ingests while queries run (the reference's consumer
+    updates state continuously); the store must stay consistent."""
+    import threading
+
+    store = ALSModelStore(device=torch.device("cpu"))
+    store.ingest([f"{i},U,1.0;2.0" for i in range(50)])
+    store.ingest(["0,I,1.0;1.0"])
+    stop = threading.Event()
+    errors = []
+
+    def writer():
+        n = 0
+        while not stop.is_set():
+            store.ingest_row(f"{n % 50},U,{float(n)};2.0")
+            n += 1
+
+    def reader():
+        while not stop.is_set():
+            try:
+                hit = store.query(f"{torch.randint(0, 50, (1,)).item()}-U")
+                assert hit is not None
+                pred = store.predict("3", "0")
+                assert pred is not None
+            except Exception as e:  # noqa: BLE001
+                errors.append(e)
+                return
+
+    threads = [threading.Thread(target=writer)] + \
+              [threading.Thread(target=reader) for _ in range(3)]
+    for t in threads:
+        t.start()
+    import time as _time
+    _time.sleep(0.5)
+    stop.set()
+    for t in threads:
+        t.join()
+    assert not errors, errors

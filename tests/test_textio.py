@@ -69,3 +69,43 @@ def test_svm_rows():
 def test_latency_csv_rows():
     assert t.als_latency_csv_row(1, 2, 3.5, 12) == "1,2,3.5,12"
     assert t.svm_latency_csv_row(0, 5, -1.0, 3) == "0,5,-1.0,3"
+
+
+# ---- property tests (hypothesis): the codecs round-trip arbitrary doubles
+
+from hypothesis import given, settings, strategies as st
+
+
+@settings(max_examples=500, deadline=None)
+@given(st.floats(allow_nan=False, allow_infinity=False))
+def test_java_double_roundtrip_property(x):
+    s = t.java_double_to_string(x)
+    assert float(s.replace("E", "e")) == x
+    # Java surface rules: decimal form iff 1e-3 <= |x| < 1e7 (or zero)
+    if x != 0 and 1e-3 <= abs(x) < 1e7:
+        assert "E" not in s and "." in s
+    elif x != 0:
+        assert "E" in s
+
+
+@settings(max_examples=200, deadline=None)
+@given(st.lists(st.floats(allow_nan=False, allow_infinity=False,
+                          width=32), min_size=1, max_size=8),
+       st.integers(min_value=0, max_value=2**31 - 1),
+       st.sampled_from(["U", "I"]))
+def test_als_row_roundtrip_property(factors, rid, kind):
+    row = t.als_factor_row(rid, kind, factors)
+    pid, pkind, pfactors = t.parse_als_row(row)
+    assert pid == str(rid) and pkind == kind
+    assert pfactors == [float(f) for f in factors]
+
+
+@settings(max_examples=200, deadline=None)
+@given(st.lists(st.tuples(st.integers(min_value=1, max_value=10**6),
+                          st.floats(allow_nan=False, allow_infinity=False)),
+                min_size=1, max_size=20))
+def test_svm_range_row_roundtrip_property(pairs):
+    row = t.svm_range_row(7, pairs)
+    b, parsed = t.parse_svm_range_row(row)
+    assert b == 7
+    assert parsed == [(i, float(w)) for i, w in pairs]

@@ -90,3 +90,24 @@ def test_bench_cpu_smoke():
                      "--rank", "8", "--ratings-per-gpu", "2000",
                      "--users-per-gpu", "150", "--items", "80"])
     assert rc == 0
+
+
+def test_bench_json_contract(capsys):
+    """The driver depends on bench.py's single JSON line: validate schema."""
+    import json
+
+    import bench
+    rc = bench.main(["--device", "cpu", "--steps", "1", "--warmup", "0",
+                     "--rank", "8", "--ratings-per-gpu", "1500",
+                     "--users-per-gpu", "100", "--items", "60"])
+    assert rc == 0
+    line = [ln for ln in capsys.readouterr().out.splitlines()
+            if ln.startswith("{")][-1]
+    out = json.loads(line)
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in out, key
+    assert out["higher_is_better"] is True and out["scaling"] == "weak"
+    assert out["data"] == "synthetic" and out["value"] > 0
+    assert out["config"]["global_batch"] == 1500 * out["n_gpus"]
