@@ -51,6 +51,11 @@ class SVMPredictBody(BaseModel):
     range: Optional[int] = None   # set -> range-partitioned lookups
 
 
+class PredictBatchBody(BaseModel):
+    users: List[str]
+    items: List[str]
+
+
 class SGDBody(BaseModel):
     ratings: List[str] = Field(description="'user\\titem\\trating' rows")
     field_delimiter: str = "\t"
@@ -178,6 +183,15 @@ def create_app(als_store: Optional[ALSModelStore] = None,
                                f"model for the query: {user},{item}"}
         return {"found": True, "prediction": pred,
                 "formatted": f"ALS Prediction =  {pred:f} "}
+
+    @app.post("/als/predict_batch")
+    def als_predict_batch(body: PredictBatchBody):
+        """Batched predictions through the device bf16 mirror + K5 kernel
+        (SURVEY.md §7 'batch queued queries into K5 launches')."""
+        if len(body.users) != len(body.items):
+            raise HTTPException(400, "users/items length mismatch")
+        preds, ok = als.predict_batch(body.users, body.items)
+        return {"predictions": preds.tolist(), "found": ok.tolist()}
 
     @app.post("/svm/predict")
     def svm_predict(body: SVMPredictBody):

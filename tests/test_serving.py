@@ -253,3 +253,13 @@ def test_store_concurrent_ingest_and_query():
     for t in threads:
         t.join()
     assert not errors, errors
+
+
+def test_rest_predict_batch(als_store, svm_store):
+    app = create_app(als_store, svm_store)
+    c = TestClient(app)
+    r = c.post("/als/predict_batch",
+               json={"users": ["1", "2", "404"],
+                     "items": ["10", "10", "10"]}).json()
+    assert r["found"] == [True, True, False]
+    assert r["predictions"][0] == pytest.approx(0.5, abs=2e-2)  # bf16 mirror
