@@ -99,6 +99,53 @@ class ALSModelStore:
         self._mirror[row] = torch.tensor(factors, dtype=torch.float32
                                          ).to(torch.bfloat16)
 
+    # ---------------------------------------------- tensor-direct attach
+
+    def attach_factors(self, user_factors: torch.Tensor,
+                       item_factors: torch.Tensor,
+                       user_ids: Optional[torch.Tensor] = None,
+                       item_ids: Optional[torch.Tensor] = None) -> None:
+        """Attach trained fp32 factor tensors directly (the in-process
+        train->serve handoff).  Payload strings are formatted LAZILY on
+        first query of a key — serving a 288-GB-scale model starts
+        instantly instead of eagerly building millions of text rows.
+        Byte-parity holds: the lazy string is the same Java-format row the
+        training job would have written and the producer re-ingested.
+        Explicitly ingested rows always take precedence (hot swap)."""
+        from ..utils.textio import format_factors  # local import cycle-safe
+        self._fmt = format_factors
+        with self._lock:
+            self._attached = {
+                "U": (user_factors.to(torch.float32).cpu(),
+                      {int(v): r for r, v in enumerate(user_ids.tolist())}
+                      if user_ids is not None else None),
+                "I": (item_factors.to(torch.float32).cpu(),
+                      {int(v): r for r, v in enumerate(item_ids.tolist())}
+                      if item_ids is not None else None),
+            }
+            if self._k is None:
+                self._k = int(user_factors.shape[1])
+            # device mirror for the batched paths
+            dev_u = user_factors.to(self.device).to(torch.bfloat16)
+            dev_v = item_factors.to(self.device).to(torch.bfloat16)
+            self._attached_dev = {"U": dev_u, "I": dev_v}
+
+    def _attached_row(self, key: str) -> Optional[List[float]]:
+        att = getattr(self, "_attached", None)
+        if att is None:
+            return None
+        try:
+            entity_id, kind = key.rsplit("-", 1)
+            fac, idmap = att.get(kind, (None, None))
+            if fac is None:
+                return None
+            row = idmap[int(entity_id)] if idmap is not None else int(entity_id)
+            if row < 0 or row >= fac.shape[0]:
+                return None
+            return fac[row].tolist()
+        except (ValueError, KeyError):
+            return None
+
     # ------------------------------------------------------------- query
 
     def query(self, key: str) -> Optional[Tuple[str, str]]:
@@ -106,11 +153,21 @@ class ALSModelStore:
         QueryClientHelper.java:135-137)."""
         with self._lock:
             payload = self._payload.get(key)
+        if payload is None:
+            vec = self._attached_row(key)
+            if vec is not None:
+                payload = self._fmt(vec)
+                with self._lock:  # cache the lazily formatted payload
+                    self._payload.setdefault(key, payload)
+                    self._vec.setdefault(key, vec)
         return None if payload is None else (key, payload)
 
     def get_vector(self, key: str) -> Optional[List[float]]:
         with self._lock:
-            return self._vec.get(key)
+            vec = self._vec.get(key)
+        if vec is None:
+            vec = self._attached_row(key)
+        return vec
 
     def predict(self, user_id: str, item_id: str) -> Optional[float]:
         """``dot(U[u], V[i])`` in fp64 from the payloads — bit-matches the
@@ -121,26 +178,55 @@ class ALSModelStore:
             return None
         return float(sum(a * b for a, b in zip(u, v)))
 
+    def _batch_rows(self, ids: List[str], kind: str
+                    ) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Resolve ids -> (mirror_row, attached_row); -1 where absent.
+        Ingested rows (the mirror) take precedence over attached tensors."""
+        att = getattr(self, "_attached", None)
+        idmap = att[kind][1] if att is not None else None
+        att_n = att[kind][0].shape[0] if att is not None else 0
+        m_rows, a_rows = [], []
+        for s in ids:
+            key = als_state_key(s, kind)
+            m = self._rows.get(key, -1)
+            a = -1
+            if m < 0 and att is not None:
+                try:
+                    a = idmap[int(s)] if idmap is not None else int(s)
+                    if a < 0 or a >= att_n:
+                        a = -1
+                except (ValueError, KeyError):
+                    a = -1
+            m_rows.append(m)
+            a_rows.append(a)
+        return (torch.tensor(m_rows, dtype=torch.int64),
+                torch.tensor(a_rows, dtype=torch.int64))
+
     def predict_batch(self, user_ids: List[str], item_ids: List[str]
                       ) -> Tuple[torch.Tensor, torch.Tensor]:
-        """Batched predictions via the K5 kernel on the device mirror.
-        Returns (predictions fp32, found mask)."""
+        """Batched predictions via the K5 kernel on the device mirror and/or
+        attached factor tensors.  Returns (predictions fp32, found mask)."""
         from .. import ops
         with self._lock:
-            u_rows = [self._rows.get(als_state_key(u, "U"), -1)
-                      for u in user_ids]
-            i_rows = [self._rows.get(als_state_key(i, "I"), -1)
-                      for i in item_ids]
+            um, ua = self._batch_rows(user_ids, "U")
+            im, ia = self._batch_rows(item_ids, "I")
             mirror = self._mirror
-        u_t = torch.tensor(u_rows, dtype=torch.int64)
-        i_t = torch.tensor(i_rows, dtype=torch.int64)
-        ok = (u_t >= 0) & (i_t >= 0)
+            att_dev = getattr(self, "_attached_dev", None)
+        ok = ((um >= 0) | (ua >= 0)) & ((im >= 0) | (ia >= 0))
         preds = torch.zeros(len(user_ids), dtype=torch.float32)
-        if mirror is not None and int(ok.sum()) > 0:
-            sel_u = u_t[ok].to(mirror.device)
-            sel_i = i_t[ok].to(mirror.device)
-            p = ops.predict_dot(mirror, mirror, sel_u, sel_i)
-            preds[ok] = p.cpu()
+        if int(ok.sum()) == 0:
+            return preds, ok
+        # group by (user source, item source); each group is one K5 launch
+        for usrc, umask in (("m", um >= 0), ("a", (um < 0) & (ua >= 0))):
+            for isrc, imask in (("m", im >= 0), ("a", (im < 0) & (ia >= 0))):
+                sel = ok & umask & imask
+                if int(sel.sum()) == 0:
+                    continue
+                U = mirror if usrc == "m" else att_dev["U"]
+                V = mirror if isrc == "m" else att_dev["I"]
+                su = (um if usrc == "m" else ua)[sel].to(U.device)
+                si = (im if isrc == "m" else ia)[sel].to(V.device)
+                preds[sel] = ops.predict_dot(U, V, su, si).cpu()
         return preds, ok
 
     # -------------------------------------------------------- online SGD

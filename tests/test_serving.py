@@ -263,3 +263,40 @@ def test_rest_predict_batch(als_store, svm_store):
                      "items": ["10", "10", "10"]}).json()
     assert r["found"] == [True, True, False]
     assert r["predictions"][0] == pytest.approx(0.5, abs=2e-2)  # bf16 mirror
+
+
+def test_attach_factors_lazy_payload():
+    """Tensor-attached serving: payloads format lazily and byte-match the
+    train->write->ingest path."""
+    import io
+
+    from flink_ms_amd.data.ratings import RatingsShape, synthetic_ratings
+    from flink_ms_amd.models.als import ALSConfig, train_als
+    u, i, r = synthetic_ratings(RatingsShape(40, 20, 600), seed=6)
+    model, _ = train_als(u, i, r, 40, 20,
+                         ALSConfig(iterations=2, num_factors=8,
+                                   lambda_=0.1, dtype=torch.float32))
+    # path A: write + ingest (the producer path)
+    uf, itf = io.StringIO(), io.StringIO()
+    model.write(uf, itf)
+    ingested = ALSModelStore(device=torch.device("cpu"))
+    ingested.ingest(uf.getvalue().splitlines())
+    ingested.ingest(itf.getvalue().splitlines())
+    # path B: attach tensors, format lazily
+    attached = ALSModelStore(device=torch.device("cpu"))
+    attached.attach_factors(model.user_factors, model.item_factors,
+                            model.user_ids, model.item_ids)
+    for key in ("0-U", "39-U", "7-I"):
+        assert attached.query(key) == ingested.query(key)
+    assert attached.query("99-U") is None
+    # predictions match
+    assert attached.predict("3", "5") == pytest.approx(
+        ingested.predict("3", "5"), rel=1e-12)
+    # ingested rows take precedence (hot swap over attached)
+    attached.ingest_row("0,U,9.0;0;0;0;0;0;0;0")
+    assert attached.query("0-U")[1].startswith("9.0")
+    # batched path over attached tensors
+    preds, ok = attached.predict_batch(["1", "99"], ["2", "2"])
+    assert ok.tolist() == [True, False]
+    exp = float(model.user_factors[1].double() @ model.item_factors[2].double())
+    assert preds[0].item() == pytest.approx(exp, abs=0.05)  # bf16 mirror
