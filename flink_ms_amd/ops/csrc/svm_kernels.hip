@@ -33,28 +33,52 @@ __global__ void k_sdca_pass(const long long* __restrict__ indptr,
     const int lane = threadIdx.x & 63;
     const long long wave = (long long)blockIdx.x * 4 + (threadIdx.x >> 6);
     const long long nwaves = (long long)gridDim.x * 4;
-    for (long long s = wave; s < nrows; s += nwaves) {
-        const long long i = perm ? (long long)perm[s] : s;
-        const long long e0 = indptr[i], e1 = indptr[i + 1];
-        const float nsq = norms_sq[i];
-        if (e1 == e0 || nsq == 0.0f) continue;
-        // margin = y_i * <v, x_i> (gather + wave reduction)
-        float part = 0.0f;
-        for (long long t = e0 + lane; t < e1; t += WAVE)
-            part += values[t] * v[indices[t]];
-        const float dot = wave_reduce_sum(part);
-        const float yi = y[i];
-        const float grad = (1.0f - yi * dot) / (nsq * scale);
-        float a = alpha[i];
-        float a_new = a + grad;
-        a_new = a_new < 0.0f ? 0.0f : (a_new > 1.0f ? 1.0f : a_new);
-        const float dalpha = a_new - a;
-        if (dalpha != 0.0f) {
-            if (lane == 0) alpha[i] = a_new;
-            const float c = dalpha * yi * scale;
-            for (long long t = e0 + lane; t < e1; t += WAVE)
-                atomicAdd(&v[indices[t]], c * values[t]);
+    if (wave >= nrows) return;
+
+    // The per-sample chain is 3-4 dependent memory round trips (perm ->
+    // indptr -> indices -> v); prefetching the NEXT sample's metadata and
+    // first 64 nonzeros while the current sample computes hides most of it
+    // (the v gather itself stays at compute time for hogwild freshness).
+    struct Meta {
+        long long i, e0, e1;
+        float nsq, yi, a, val;
+        int idx;
+    };
+    auto fetch = [&](long long s, Meta& m) {
+        if (s >= nrows) { m.e0 = m.e1 = 0; m.idx = -1; return; }
+        m.i = perm ? (long long)perm[s] : s;
+        m.e0 = indptr[m.i];
+        m.e1 = indptr[m.i + 1];
+        m.nsq = norms_sq[m.i];
+        m.yi = y[m.i];
+        m.a = alpha[m.i];
+        const long long t = m.e0 + lane;
+        m.idx = (t < m.e1) ? indices[t] : -1;
+        m.val = (t < m.e1) ? values[t] : 0.0f;
+    };
+
+    Meta cur, nxt;
+    fetch(wave, cur);
+    for (long long s = wave; s < nrows; s = s + nwaves) {
+        fetch(s + nwaves, nxt);   // in flight across the current compute
+        if (cur.e1 > cur.e0 && cur.nsq != 0.0f) {
+            float part = (cur.idx >= 0) ? cur.val * v[cur.idx] : 0.0f;
+            for (long long t = cur.e0 + 64 + lane; t < cur.e1; t += WAVE)
+                part += values[t] * v[indices[t]];
+            const float dot = wave_reduce_sum(part);
+            const float grad = (1.0f - cur.yi * dot) / (cur.nsq * scale);
+            float a_new = cur.a + grad;
+            a_new = a_new < 0.0f ? 0.0f : (a_new > 1.0f ? 1.0f : a_new);
+            const float dalpha = a_new - cur.a;
+            if (dalpha != 0.0f) {
+                if (lane == 0) alpha[cur.i] = a_new;
+                const float c = dalpha * cur.yi * scale;
+                if (cur.idx >= 0) atomicAdd(&v[cur.idx], c * cur.val);
+                for (long long t = cur.e0 + 64 + lane; t < cur.e1; t += WAVE)
+                    atomicAdd(&v[indices[t]], c * values[t]);
+            }
         }
+        cur = nxt;
     }
 }
 
