@@ -27,6 +27,7 @@ import torch
 from .. import ops
 from ..data.blocked import CSR, csr_from_coo
 from ..parallel.dist import DistContext, get_context
+from ..parallel.routing import plan_exchange
 from ..parallel.shard import Partition, allgather_rows, exchange_ratings_by_owner
 from ..utils.textio import als_factor_row
 
@@ -39,6 +40,9 @@ class ALSConfig:
     seed: int = 42
     # compute dtype of the factor operands fed to the Gramian kernels
     dtype: torch.dtype = torch.bfloat16
+    # C1 exchange: 'auto' picks routed all-to-all-v when the referenced
+    # fraction of the opposite side is sparse, full all-gather otherwise
+    routed_exchange: str = 'auto'
 
 
 @dataclass
@@ -110,6 +114,17 @@ class ALSTrainer:
             (ik.long() - ilo).to(torch.int32), io.to(torch.int32), iv,
             num_rows=max(ihi - ilo, 1), num_cols=num_users).to(dev)
 
+        # C1 plan: routed all-to-all-v (flink-ml routing tables) or full
+        # all-gather per side; routed remaps CSR columns to compact rows
+        self.item_route, rem = plan_exchange(
+            ctx, self.ipart, self.user_csr.indices, self.cfg.routed_exchange)
+        if self.item_route is not None:
+            self.user_csr.indices = rem.to(dev)
+        self.user_route, rem = plan_exchange(
+            ctx, self.upart, self.item_csr.indices, self.cfg.routed_exchange)
+        if self.user_route is not None:
+            self.item_csr.indices = rem.to(dev)
+
         # degree-descending schedule so heavy entities launch first
         self.user_order = torch.argsort(
             self.user_csr.row_counts(), descending=True).to(torch.int32).to(dev)
@@ -137,8 +152,11 @@ class ALSTrainer:
         seconds (max over ranks)."""
         ctx = self.ctx
         t0 = time.perf_counter()
-        # C1: item factors to every rank, then solve local users
-        item_full = allgather_rows(ctx, self.item_shard, self.num_items)
+        # C1: item factors to every rank (routed a2a-v or all-gather),
+        # then solve local users
+        item_full = (self.item_route.exchange(self.item_shard)
+                     if self.item_route is not None
+                     else allgather_rows(ctx, self.item_shard, self.num_items))
         self.user_f32 = ops.als_solve_side(
             self.user_csr, item_full, self.cfg.lambda_,
             out_bf16=self.user_shard[: self.user_csr.num_rows]
@@ -148,7 +166,9 @@ class ALSTrainer:
             self.user_shard[: self.user_csr.num_rows, : self.cfg.num_factors] = (
                 self.user_f32.to(self.cfg.dtype))
         # C1': user factors to every rank, then solve local items
-        user_full = allgather_rows(ctx, self.user_shard, self.num_users)
+        user_full = (self.user_route.exchange(self.user_shard)
+                     if self.user_route is not None
+                     else allgather_rows(ctx, self.user_shard, self.num_users))
         self.item_f32 = ops.als_solve_side(
             self.item_csr, user_full, self.cfg.lambda_,
             out_bf16=self.item_shard[: self.item_csr.num_rows]

@@ -43,7 +43,7 @@ def _init(rank, world, port):
     return D.init_from_env(backend="gloo")
 
 
-def _als_worker(rank, world, port, q):
+def _als_worker(rank, world, port, q, mode="auto"):
     torch.manual_seed(0)
     ctx = _init(rank, world, port)
     from flink_ms_amd.data.ratings import RatingsShape, synthetic_ratings
@@ -53,7 +53,7 @@ def _als_worker(rank, world, port, q):
     # split triples arbitrarily (by parity) — setup() re-exchanges by owner
     mask = torch.arange(shape.num_ratings) % world == rank
     cfg = ALSConfig(iterations=3, num_factors=8, lambda_=0.2,
-                    dtype=torch.float32)
+                    dtype=torch.float32, routed_exchange=mode)
     tr = ALSTrainer(cfg, ctx)
     tr.setup(u[mask].long(), i[mask].long(), r[mask],
              shape.num_users, shape.num_items)
@@ -69,8 +69,9 @@ def _als_worker(rank, world, port, q):
 
 
 @pytest.mark.timeout(300)
-def test_distributed_als_matches_single_process():
-    results = _run_workers(_als_worker)
+@pytest.mark.parametrize("mode", ["off", "on"])
+def test_distributed_als_matches_single_process(mode):
+    results = _run_workers(_als_worker, extra=(mode,))
     # single-process reference on the full triple set
     from flink_ms_amd.data.ratings import RatingsShape, synthetic_ratings
     from flink_ms_amd.models.als import ALSConfig, train_als
