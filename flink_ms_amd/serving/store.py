@@ -50,6 +50,9 @@ class ALSModelStore:
         self._mirror: Optional[torch.Tensor] = None  # [cap, k] bf16
         self._mirror_len = 0
         self._k: Optional[int] = None
+        self._attached = None      # tensor-direct factors (attach_factors)
+        self._attached_dev = None
+        self._fmt = None
 
     # ------------------------------------------------------------ ingest
 

@@ -165,3 +165,37 @@ def test_generator_clis(tmp_path):
     assert svm_model_generator.main(["--numFeatures", "50", "--range", "10",
                                      "--output", str(out2)]) == 0
     assert len(open(out2).read().splitlines()) == 5
+
+
+def test_interactive_clients(tmp_path, server, capsys, monkeypatch):
+    """REPL clients (ALSPredict / SVMPredict parity) against the live server."""
+    import io as io_mod
+
+    from flink_ms_amd.cli import als_predict, svm_predict
+    port = str(server)
+    # seed some state
+    assert producer.main(["--input", _write(tmp_path / "m1", "7,U,1.0;2.0\n8,I,0.5;0.25\n"),
+                          "--model", "als", "--server", "127.0.0.1",
+                          "--port", port]) == 0
+    monkeypatch.setattr("sys.stdin", io_mod.StringIO("7,8\n7,999\n"))
+    assert als_predict.main(["job0", "127.0.0.1", port]) == 0
+    out = capsys.readouterr().out
+    assert "ALS Prediction =  1.000000" in out      # 1*0.5 + 2*0.25
+    assert "do not exist in the model for the query: 7,999" in out
+    # svm
+    # fresh ids (the module-scoped server is shared with other tests)
+    assert producer.main(["--input", _write(tmp_path / "m2", "900001,2.0\n"),
+                          "--model", "svm", "--server", "127.0.0.1",
+                          "--port", port]) == 0
+    monkeypatch.setattr("sys.stdin",
+                        io_mod.StringIO("900001:3.0 900005:1.0\n"))
+    assert svm_predict.main(["job0", "127.0.0.1", port, "true"]) == 0
+    out = capsys.readouterr().out
+    assert "Could not find the value for feature ID: 900005" in out
+    assert "SVM Prediction =  6.000000" in out
+
+
+def _write(path, content):
+    with open(path, "w") as f:
+        f.write(content)
+    return str(path)
