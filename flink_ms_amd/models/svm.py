@@ -37,6 +37,9 @@ class SVMConfig:
     regularization: float = 0.01   # lambda
     stepsize: float = 1.0
     seed: int = 42
+    # capture the local_iterations SDCA passes in one hipGraph (HIP graphs
+    # replace launch overhead for the launch-dense inner loop; GPU only)
+    use_graphs: bool = True
 
 
 @dataclass
@@ -97,18 +100,57 @@ class SVMTrainer:
         # so per-pass reshuffles buy nothing but a host randperm + H2D copy
         self._perm = torch.randperm(
             local_csr.num_rows, generator=self._gen).to(torch.int32).to(dev)
+        self._graph = None
+        self._v_static: Optional[torch.Tensor] = None
+        if dev.type == "cuda" and self.cfg.use_graphs:
+            self._capture_local_solver()
+
+    def _capture_local_solver(self) -> None:
+        """Capture the local_iterations SDCA passes into one hipGraph
+        (torch.cuda.CUDAGraph is hipGraph on ROCm).  alpha and the static
+        primal image are the graph's fixed buffers; each outer iteration
+        re-seeds v from w and replays."""
+        self._v_static = torch.zeros_like(self.w)
+
+        def passes():
+            for _ in range(self.cfg.local_iterations):
+                ops.sdca_pass(self.csr, self.y, self.alpha, self._v_static,
+                              self.cfg.regularization, self.n_global,
+                              norms_sq=self.norms_sq, perm=self._perm)
+
+        # warmup on a side stream (graph-capture requirement)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            a_save = self.alpha.clone()
+            passes()
+            self.alpha.copy_(a_save)
+        torch.cuda.current_stream().wait_stream(s)
+        try:
+            self._graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self._graph):
+                passes()
+            self.alpha.copy_(a_save)
+        except Exception:   # capture unsupported -> eager fallback
+            self._graph = None
 
     def step(self) -> float:
         """One outer CoCoA iteration; returns wall seconds (max over ranks)."""
         ctx = self.ctx
         cfg = self.cfg
         t0 = time.perf_counter()
-        v = self.w.clone()
-        a0 = self.alpha.clone()
-        for _ in range(cfg.local_iterations):
-            ops.sdca_pass(self.csr, self.y, self.alpha, v,
-                          cfg.regularization, self.n_global,
-                          norms_sq=self.norms_sq, perm=self._perm)
+        if self._graph is not None:
+            v = self._v_static
+            v.copy_(self.w)
+            a0 = self.alpha.clone()
+            self._graph.replay()
+        else:
+            v = self.w.clone()
+            a0 = self.alpha.clone()
+            for _ in range(cfg.local_iterations):
+                ops.sdca_pass(self.csr, self.y, self.alpha, v,
+                              cfg.regularization, self.n_global,
+                              norms_sq=self.norms_sq, perm=self._perm)
         K = max(ctx.world_size, 1)
         dw = v - self.w
         ctx.all_reduce_(dw)

@@ -184,3 +184,24 @@ def test_slabbed_solve_matches_unslabbed(gpu):
     slabbed = ops.als_solve_side(csr, fac, reg=0.4, slab_rows=77)
     torch.cuda.synchronize()
     assert torch.equal(full, slabbed)
+
+
+def test_svm_graph_capture_matches_eager(gpu):
+    """hipGraph-captured local solver must train like the eager path."""
+    from flink_ms_amd.data.libsvm import LibSVMShape, synthetic_libsvm
+    from flink_ms_amd.models.svm import SVMConfig, SVMTrainer
+    csr, y = synthetic_libsvm(LibSVMShape(8000, 600, 16), seed=12,
+                              separable=True)
+    objs = {}
+    for graphs in (False, True):
+        tr = SVMTrainer(SVMConfig(iterations=4, local_iterations=3,
+                                  regularization=0.01, use_graphs=graphs))
+        tr.ctx.device = gpu
+        tr.setup(csr, y)
+        if graphs:
+            assert tr._graph is not None, "graph capture failed"
+        tr.fit()
+        objs[graphs] = tr.objective()
+    # hogwild nondeterminism aside, both must converge comparably
+    assert objs[True] < 0.5 and objs[False] < 0.5
+    assert abs(objs[True] - objs[False]) < 0.15
