@@ -1,36 +1,26 @@
-"""In-tree build of the flink_ms_amd HIP extension for MI355X (gfx950).
-
-    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
-
-The built .so lands next to the package (flink_ms_amd/_hip_ops*.so) so it
-travels with the source tree to GPU boxes.
+"""Package setup.  The HIP extension is built by a direct hipcc driver
+(`python -m flink_ms_amd.ops.build`), NOT by torch's CUDAExtension pipeline
+(whose CUDA->HIP rewriter would mangle the native HIP sources); the custom
+build_ext below just delegates so `python setup.py build_ext --inplace`
+keeps working.
 """
-import os
 
 from setuptools import setup
+from setuptools.command.build_ext import build_ext as _build_ext
 
-os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
-from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+class HipccBuildExt(_build_ext):
+    def run(self):
+        from flink_ms_amd.ops.build import build
+        build(verbose=True)
+
 
 setup(
     name="flink_ms_amd",
     version="0.1.0",
-    packages=["flink_ms_amd"],
-    ext_modules=[
-        CUDAExtension(
-            name="flink_ms_amd._hip_ops",
-            sources=[
-                "flink_ms_amd/ops/csrc/bindings.cpp",
-                "flink_ms_amd/ops/csrc/als_kernels.hip",
-                "flink_ms_amd/ops/csrc/svm_kernels.hip",
-                "flink_ms_amd/ops/csrc/serve_kernels.hip",
-            ],
-            extra_compile_args={
-                "cxx": ["-O2"],
-                "nvcc": ["-O3", "-std=c++17"],
-            },
-        )
-    ],
-    cmdclass={"build_ext": BuildExtension},
+    packages=["flink_ms_amd", "flink_ms_amd.cli", "flink_ms_amd.data",
+              "flink_ms_amd.models", "flink_ms_amd.ops",
+              "flink_ms_amd.parallel", "flink_ms_amd.serving",
+              "flink_ms_amd.utils"],
+    cmdclass={"build_ext": HipccBuildExt},
 )
