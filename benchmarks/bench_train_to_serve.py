@@ -63,8 +63,9 @@ def main():
             time.sleep(0.1)
     # spot check: served prediction == trained model dot
     resp = client.als_predict("0", "0")
-    exp = float(model.user_factors[0].double()
-                @ model.item_factors[0].double())
+    exp = float(model.user_factors[0].double().cpu()
+                @ model.item_factors[0].double().cpu())
+    assert resp.get("found"), resp
     assert abs(resp["prediction"] - exp) < 1e-9, (resp, exp)
     res = als_predict_random(num_queries=2000,
                              upper_user_id=ML25M_SHAPE.num_users - 1,

@@ -95,8 +95,11 @@ def create_app(als_store: Optional[ALSModelStore] = None,
                checkpoint_data_uri: Optional[str] = None,
                checkpoint_interval_ms: int = 60000) -> FastAPI:
     app = FastAPI(title="flink_ms_amd model serving")
-    als = als_store or ALSModelStore()
-    svm = svm_store or SVMModelStore()
+    # NOTE: explicit None checks — an EMPTY store is falsy (__len__ == 0),
+    # so `als_store or ALSModelStore()` would silently drop a store that is
+    # populated later (e.g. via attach_factors)
+    als = als_store if als_store is not None else ALSModelStore()
+    svm = svm_store if svm_store is not None else SVMModelStore()
     app.state.als = als
     app.state.svm = svm
     app.state.checkpoint_uri = checkpoint_data_uri
