@@ -35,9 +35,25 @@ def build_app(params: Params):
 
 def main(argv=None) -> int:
     params = Params.from_args(sys.argv[1:] if argv is None else argv)
-    app = build_app(params)
-    uvicorn.run(app, host=params.get("host", "0.0.0.0"),
-                port=params.get_int("port", 6123), log_level="warning")
+    # fixed-delay restart strategy (consumer parity: 3 attempts / 10 s,
+    # ALSKafkaConsumer.java:48-51); state survives restarts via the
+    # checkpoint snapshots (restore by passing the snapshot as --alsModel)
+    attempts = params.get_int("restartAttempts", 3)
+    delay_s = params.get_int("restartDelay", 10000) / 1000.0
+    import time
+    for attempt in range(attempts + 1):
+        try:
+            app = build_app(params)
+            uvicorn.run(app, host=params.get("host", "0.0.0.0"),
+                        port=params.get_int("port", 6123),
+                        log_level="warning")
+            return 0
+        except Exception as e:  # noqa: BLE001
+            if attempt == attempts:
+                raise
+            print(f"serving job failed ({e!r}); restart "
+                  f"{attempt + 1}/{attempts} in {delay_s}s")
+            time.sleep(delay_s)
     return 0
 
 

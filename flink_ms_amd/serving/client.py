@@ -49,11 +49,21 @@ class QueryClientHelper:
         r.raise_for_status()
         return r.json()
 
-    def ingest_rows(self, model: str, rows: List[str]) -> int:
-        r = self._client.post(f"{self.base}/model/{model}/rows",
-                              json={"rows": rows})
-        r.raise_for_status()
-        return r.json()["ingested"]
+    def ingest_rows(self, model: str, rows: List[str],
+                    retries: int = 3) -> int:
+        """At-least-once publish (producer parity: setFlushOnCheckpoint,
+        ALSKafkaProducer.java:35-37): retry on transport failure; ingest is
+        idempotent (last writer wins per key)."""
+        last: Optional[Exception] = None
+        for _ in range(retries + 1):
+            try:
+                r = self._client.post(f"{self.base}/model/{model}/rows",
+                                      json={"rows": rows})
+                r.raise_for_status()
+                return r.json()["ingested"]
+            except httpx.TransportError as e:
+                last = e
+        raise last
 
     def sgd_update(self, ratings: List[str], **kw) -> dict:
         payload = {"ratings": ratings}
