@@ -2,14 +2,16 @@
 
 Replaces the reference's training pipeline (flink-als/src/main/scala/de/tub/
 it4bi/ALSImpl.scala:35-63 driving flink-ml's blocked ALS) with a PyTorch-ROCm
-loop over the fused HIP normal-equation kernels:
+loop over the HIP normal-equation kernels:
 
   per iteration (reference: Flink bulk iteration):
-    1. solve user factors from item factors   (K1+K2 fused kernel)
+    1. solve user factors from item factors   (K1 MFMA Gramian kernel +
+       K2 batched LDL solve; fused single-kernel variant for k > 64)
     2. solve item factors from user factors
-  multi-GPU: factors are range-sharded; each half-iteration all-gathers the
-  opposite side's bf16 shard replica over xGMI (C1) instead of flink-ml's
-  per-block routed shuffle; ratings are exchanged once at setup.
+  multi-GPU: factors are range-sharded; each half-iteration exchanges the
+  opposite side's bf16 factors over xGMI (C1) — routed all-to-all-v from
+  precomputed request tables, or a bucketed all-gather in dense regimes —
+  and ratings are routed to their owners once at setup.
 
 Flag parity: iterations(10), numFactors(10), lambda(0.9),
 seed(42), blocks (block count maps to GPU count / is advisory here).
