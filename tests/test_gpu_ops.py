@@ -206,3 +206,39 @@ def test_svm_graph_capture_matches_eager(gpu):
     # (objective starts at 1.0 on this shape)
     assert objs[True] < 0.7 and objs[False] < 0.7
     assert abs(objs[True] - objs[False]) < 0.15
+
+
+def test_gramian_heavy_row_skew(gpu):
+    """A pathological hot entity (100K ratings on one row, reference-world:
+    a blockbuster item) must stay correct through the long chunk loop."""
+    g = torch.Generator().manual_seed(41)
+    hot = 100_000
+    rows = torch.cat([torch.zeros(hot, dtype=torch.int32),
+                      torch.randint(1, 64, (5_000,), generator=g,
+                                    dtype=torch.int32)])
+    cols = torch.randint(0, 200, (hot + 5_000,), generator=g,
+                         dtype=torch.int32)
+    vals = torch.rand(hot + 5_000, generator=g) * 4.5 + 0.5
+    csr = csr_from_coo(rows, cols, vals, 64, 200).to(gpu)
+    fac = (torch.randn(200, 32, generator=g) * 0.3).to(torch.bfloat16)
+    out = ops.als_solve_side(csr, fac.to(gpu), reg=0.9)
+    ref = R.als_solve_side_reference(csr.to("cpu"), fac.to(torch.float32),
+                                     reg=0.9)
+    torch.cuda.synchronize()
+    err = (out.cpu() - ref).abs().amax() / ref.abs().amax()
+    assert err < 2e-2, f"heavy-row rel err {err}"
+
+
+def test_solve_rank10_padding(gpu):
+    """The reference's default numFactors=10 exercises the pad-to-16 path
+    (padded diag regularized, padded solution sliced away)."""
+    csr = _rand_csr(rows=300, cols=150, nnz=9_000, seed=51, device=gpu)
+    fac = (torch.randn(150, 10, generator=torch.Generator().manual_seed(52))
+           * 0.5).to(torch.bfloat16)
+    out = ops.als_solve_side(csr, fac.to(gpu), reg=0.5)
+    assert out.shape == (300, 10)
+    ref = R.als_solve_side_reference(csr.to("cpu"), fac.to(torch.float32),
+                                     reg=0.5)
+    torch.cuda.synchronize()
+    err = (out.cpu() - ref).abs().amax() / ref.abs().amax()
+    assert err < 2e-2, f"rank-10 rel err {err}"
