@@ -1,10 +1,12 @@
 """Mean-vector job (reference flink-als/.../ALSMeanVector.scala rebuild).
 
-Flags: --type item|user (required), --input, --output.
+Flags: --type item|user (required), --input (file OR a directory of part
+files, as distributed training writes), --output.
 """
 import sys
 
 from ..models.mean_vector import mean_vector_rows
+from ..serving.app import _read_rows
 from ..utils.params import Params
 
 
@@ -17,8 +19,8 @@ def main(argv=None) -> int:
         factor_type = "U"
     else:
         raise ValueError("specify type as either 'item' or 'user'.")
-    with open(params.get_required("input")) as f:
-        row = mean_vector_rows(f, factor_type)
+    row = mean_vector_rows(_read_rows(params.get_required("input")),
+                           factor_type)
     if params.has("output"):
         with open(params.get("output"), "w") as f:
             f.write(row + "\n")

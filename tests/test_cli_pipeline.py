@@ -199,3 +199,38 @@ def _write(path, content):
     with open(path, "w") as f:
         f.write(content)
     return str(path)
+
+
+@pytest.mark.timeout(300)
+def test_distributed_cli_training(tmp_path):
+    """torchrun 2-rank als_train writes part-file directories (Flink
+    writeAsText parity) consumable by the mean-vector job and producer."""
+    import subprocess
+    import sys
+
+    shape = RatingsShape(80, 40, 1500)
+    u, i, r = synthetic_ratings(shape, seed=3)
+    csv = tmp_path / "ratings.csv"
+    with open(csv, "w") as f:
+        f.write("u,i,r\n")
+        for a, b, c in zip(u.tolist(), i.tolist(), r.tolist()):
+            f.write(f"{a},{b},{c}\n")
+    uf, if_ = tmp_path / "uf", tmp_path / "if"
+    res = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29778", "-m", "flink_ms_amd.cli.als_train",
+         "--input", str(csv), "--iterations", "2", "--numFactors", "8",
+         "--lambda", "0.1", "--userFactors", str(uf),
+         "--itemFactors", str(if_)],
+        capture_output=True, text=True, timeout=240)
+    assert res.returncode == 0, res.stderr[-2000:]
+    parts = sorted(p.name for p in uf.iterdir())
+    assert parts == ["part-0", "part-1"]
+    rows = sum(1 for p in uf.iterdir() for _ in open(p))
+    assert rows == 80
+    # mean-vector job over the part directory
+    out = tmp_path / "umean"
+    assert als_mean_vector.main(["--type", "user", "--input", str(uf),
+                                 "--output", str(out)]) == 0
+    assert open(out).read().startswith("MEAN,U,")
