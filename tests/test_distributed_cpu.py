@@ -140,3 +140,25 @@ def test_distributed_cocoa_svm():
     margins = R.svm_margins_reference(csr, w0)
     acc = float((margins.sign() == y).float().mean())
     assert acc > 0.85, acc
+
+
+@pytest.mark.timeout(300)
+def test_distributed_als_world4_routed():
+    """4-rank gloo run with the routed all-to-all-v exchange forced on:
+    wider routing topology than the world-2 cases."""
+    ctx = mp.get_context("spawn")
+    port = 29500 + ((os.getpid() + 7) % 500)
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_als_worker, args=(rank, 4, port, q, "on"))
+             for rank in range(4)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(4):
+        rank, payload = q.get(timeout=180)
+        results[rank] = payload
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    ids = [i for r in sorted(results) for i in results[r]["user_ids"]]
+    assert sorted(ids) == list(range(120))  # every user solved exactly once
