@@ -143,6 +143,7 @@ __global__ void k_mfma_probe_f32(const float* __restrict__ Amat,  // [16][4]
 __global__ void k_mfma_probe_bf16(const unsigned short* __restrict__ Xt,
                                   const unsigned short* __restrict__ Yt,
                                   float* __restrict__ C) {
+    auto stage_xor = [](int row) -> unsigned { return (row & 8) ? 32u : 0u; };
     constexpr int SP = 32;  // 2 tiles of 16 cols; SP*2 = 64B, XOR-closed
     __shared__ __align__(16) unsigned short st[32 * SP];
     const int lane = threadIdx.x;  // launched with 64 threads

@@ -7,25 +7,26 @@
 // re-instantiate the geometry/device functions
 #include "als_kernels_device.inc"
 
-// Dump the raw staged LDS bytes (after one stage_chunk) as bf16 elements,
-// un-XOR-ed back to logical [32][SP] layout.
+// Dump the staged Gt LDS image (one chunk): out[(K+16)][32] bf16.
 template <int KT>
 __launch_bounds__(256)
 __global__ void k_stage_dump(const long long* __restrict__ indptr,
                              const int* __restrict__ indices,
                              const float* __restrict__ values,
                              const unsigned short* __restrict__ factors,
-                             unsigned short* __restrict__ out /*[32][SP]*/) {
-    constexpr int SP = Geo<KT>::SP;
+                             unsigned short* __restrict__ out) {
+    constexpr int K = Geo<KT>::K;
+    constexpr int TROW = Geo<KT>::TROW;
     __shared__ __align__(16) char smem[Geo<KT>::SMEM];
     const long long p0 = indptr[0];
     const int n = (int)(indptr[1] - p0);
+    stage_zero_pad<KT>(smem);
     stage_chunk<KT>(smem, indices, values, factors, p0, n);
     __syncthreads();
-    for (int i = threadIdx.x; i < 32 * SP; i += 256) {
-        const int row = i / SP, col = i % SP;
-        unsigned byte = (unsigned)(row * SP + col) * 2u ^ stage_xor(row);
-        out[i] = *(const unsigned short*)(smem + byte);
+    for (int i = threadIdx.x; i < (K + 16) * 32; i += 256) {
+        const int row = i / 32, col = i % 32;
+        out[i] = *(const unsigned short*)(smem + (long long)row * TROW
+                                          + col * 2);
     }
 }
 
@@ -40,6 +41,7 @@ __global__ void k_frag_dump(const long long* __restrict__ indptr,
     __shared__ __align__(16) char smem[Geo<KT>::SMEM];
     const long long p0 = indptr[0];
     const int n = (int)(indptr[1] - p0);
+    stage_zero_pad<KT>(smem);
     stage_chunk<KT>(smem, indices, values, factors, p0, n);
     __syncthreads();
     const int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
