@@ -31,7 +31,10 @@ from ..data.blocked import CSR, csr_from_coo
 from ..parallel.dist import DistContext, get_context
 from ..parallel.routing import plan_exchange
 from ..parallel.shard import Partition, allgather_rows, exchange_ratings_by_owner
+from ..utils.logging import get_logger
 from ..utils.textio import als_factor_row
+
+log = get_logger("flink_ms_amd.als")
 
 
 @dataclass
@@ -146,6 +149,12 @@ class ALSTrainer:
                                       dtype=self.cfg.dtype, device=dev)
         self.user_f32: Optional[torch.Tensor] = None
         self.item_f32: Optional[torch.Tensor] = None
+        log.info(
+            "ALS setup: %d users x %d items, %d local ratings, k=%d, "
+            "world=%d, exchange=%s",
+            num_users, num_items, self.user_csr.nnz, k, ctx.world_size,
+            "routed-a2av" if self.item_route or self.user_route
+            else "allgather")
 
     # -- iteration -------------------------------------------------------
 
@@ -183,6 +192,8 @@ class ALSTrainer:
             torch.cuda.synchronize()
         dt = ctx.max_scalar(time.perf_counter() - t0)
         self.timings.iter_seconds.append(dt)
+        log.debug("ALS iteration %d: %.3f ms",
+                  len(self.timings.iter_seconds), dt * 1e3)
         return dt
 
     def fit(self) -> ALSModel:

@@ -26,7 +26,10 @@ import torch
 from .. import ops
 from ..data.blocked import CSR
 from ..parallel.dist import DistContext, get_context
+from ..utils.logging import get_logger
 from ..utils.textio import svm_bucket_of, svm_flat_row, svm_range_row
+
+log = get_logger("flink_ms_amd.svm")
 
 
 @dataclass
@@ -104,6 +107,9 @@ class SVMTrainer:
         self._v_static: Optional[torch.Tensor] = None
         if dev.type == "cuda" and self.cfg.use_graphs:
             self._capture_local_solver()
+        log.info("CoCoA setup: %d local x %d global samples, %d features, "
+                 "graphs=%s", local_csr.num_rows, self.n_global, self.d,
+                 self._graph is not None)
 
     def _capture_local_solver(self) -> None:
         """Capture the local_iterations SDCA passes into one hipGraph
