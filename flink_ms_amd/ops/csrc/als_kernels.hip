@@ -13,21 +13,26 @@
 //          an extra staged 16-column block as a bf16 hi/lo pair, so b falls
 //          out of the same MFMAs (columns 0/1 of the EXT tiles).
 //   K2  p_u = A_u^{-1} b_u
-//       -> fused in-LDS Cholesky factorization + wave-level triangular
-//          solves, so A never round-trips through HBM.
+//       -> in-LDS LDL^T factorization (register panels + f32-MFMA trailing
+//          updates) and substitution solves: wave-per-entity for k <= 64
+//          (k_ldl_solve_wave, the modular default) or 256-thread blocks up
+//          to k = 128 (cholesky_lds/solve_lds_block, also fused with K1 in
+//          k_als_solve_fused so A never round-trips through HBM).
 //
-// One workgroup (256 threads = 4 waves) per entity; grid = #entities.
-// Wave w owns accumulator tiles t = w, w+4, w+8, ... of the upper-triangle +
-// EXT tile list.  Factor rank k = 16*KT, KT in 1..8 (wrappers pad).
+// Gramian geometry: one workgroup (256 threads = 4 waves) per entity;
+// grid = #entities; wave w owns accumulator tiles t = w, w+4, ... of the
+// upper-triangle + EXT tile list.  Factor rank k = 16*KT, KT 1..8
+// (wrappers pad).
 //
-// LDS layout per block (union; stage dies before A is born):
-//   stage: [32][SP] bf16 rows (SP = roundup(K+32,32)), columns 0..K-1 the
-//          gathered factor row, K..K+15 the [r_hi, r_lo, 0...] rating block.
-//          Byte addresses are XOR'd by 32 on rows with bit 3 set so the two
-//          16-lane halves of a b16 lane-group land on disjoint banks
-//          (cdna_hip_programming.md §6 Guideline 4).
+// LDS layout per block (union; the stage dies before A is born):
+//   stage: TRANSPOSED Gt[K+16 rows][32 ratings] bf16, 96-B row stride
+//          (24 dwords: 16 used + 8 pad).  Rows 0..K-1 hold the gathered
+//          factor columns, rows K..K+1 the rating hi/lo pair, K+2..K+15
+//          zeros.  Each MFMA fragment is ONE bank-conflict-free
+//          ds_read_b128; staging transposes 4x4 bf16 blocks in registers
+//          (als_kernels_device.inc).
 //   A:     [K][K+1] fp32 (padded leading dim -> conflict-free column walks)
-//   b_hi/b_lo: [K] fp32 each, after A.
+//   b_hi/b_lo: [K] fp32 each, then a 16x16 pivot-column scratch.
 
 #include "common.hip.h"
 
