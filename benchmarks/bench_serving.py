@@ -92,6 +92,88 @@ def fixed_qps_als(port, qps, duration_s=5.0, workers=8,
             "p99_ms": pct(0.99)}
 
 
+def sharded_qps(als_rows, n_shards=4, qps=4000, duration_s=5.0,
+                workers=16):
+    """Key-partitioned serving: N shard processes + routed clients.
+    Measures aggregate sustained QPS + latency at a fixed offered rate."""
+    import multiprocessing as mp
+    import queue as queue_mod
+    import random
+    import threading as th
+
+    from flink_ms_amd.cli.serve import _run_shard
+    from flink_ms_amd.serving.sharding import ShardedQueryClient
+
+    base = None
+    socks = []
+    ports = []
+    for _ in range(n_shards):
+        s = socket.socket(); s.bind(("127.0.0.1", 0))
+        ports.append(s.getsockname()[1]); socks.append(s)
+    for s in socks:
+        s.close()
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_run_shard, args=({}, p), daemon=True)
+             for p in ports]
+    for p in procs:
+        p.start()
+    probe = ShardedQueryClient([("127.0.0.1", p) for p in ports], 2.0)
+    for c in probe.clients:
+        for _ in range(200):
+            try:
+                c._client.get(c.base + "/healthz").raise_for_status()
+                break
+            except Exception:
+                time.sleep(0.1)
+    probe.ingest_rows("als", als_rows)
+    probe.close()
+
+    rng = random.Random(11)
+    n = int(qps * duration_s)
+    t0 = time.perf_counter() + 0.3
+    jobs = queue_mod.Queue()
+    ids = [r.split(",", 1)[0] for r in als_rows if ",U," in r]
+    iids = [r.split(",", 1)[0] for r in als_rows if ",I," in r]
+    for k in range(n):
+        jobs.put((t0 + k / qps, rng.choice(ids), rng.choice(iids)))
+    lats = []
+    lock = th.Lock()
+
+    def worker():
+        cl = ShardedQueryClient([("127.0.0.1", p) for p in ports], 5.0)
+        while True:
+            try:
+                due, u, i = jobs.get_nowait()
+            except queue_mod.Empty:
+                cl.close()
+                return
+            now = time.perf_counter()
+            if due > now:
+                time.sleep(due - now)
+            s0 = time.perf_counter()
+            r = cl.als_predict(u, i)
+            assert r["found"]
+            with lock:
+                lats.append((time.perf_counter() - s0) * 1000.0)
+
+    threads = [th.Thread(target=worker) for _ in range(workers)]
+    start = time.perf_counter()
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    wall = time.perf_counter() - start
+    for p in procs:
+        p.terminate()
+    lats.sort()
+
+    def pct(x):
+        return lats[min(len(lats) - 1, int(x * len(lats)))] if lats else None
+    return {"shards": n_shards, "target_qps": qps,
+            "achieved_qps": len(lats) / wall, "p50_ms": pct(0.5),
+            "p95_ms": pct(0.95)}
+
+
 def main():
     t0 = time.perf_counter()
     als, svm = build_stores()
@@ -133,6 +215,10 @@ def main():
     for qps in (500, 2000):
         results[f"als_http_qps{qps}"] = fixed_qps_als(port, qps)
     srv.should_exit = True
+
+    # key-partitioned scale-out (4 shard processes, routed clients)
+    rows = als.snapshot_rows()
+    results["als_http_sharded4"] = sharded_qps(rows, n_shards=4, qps=6000)
     print(json.dumps(results, indent=1), flush=True)
 
 

@@ -6,6 +6,11 @@ surface) with periodic checkpoints.  Flags mirror the consumers
 (rocksdb|fs|memory -- accepted for parity; the state lives in process
 memory + model-format snapshots), --checkPointInterval (60000), plus
 --port (6123), --host, --alsModel/--svmModel (preload model files).
+
+``--shards N`` key-partitions the state across N serving processes on
+ports port..port+N-1 (the reference shards keyed state across
+TaskManagers); clients and producers route by
+``serving.sharding.shard_of`` (ShardedQueryClient).
 """
 import sys
 
@@ -33,8 +38,30 @@ def build_app(params: Params):
     )
 
 
+def _run_shard(params_dict, port):
+    params = Params(params_dict)
+    app = build_app(params)
+    uvicorn.run(app, host=params.get("host", "0.0.0.0"), port=port,
+                log_level="warning")
+
+
 def main(argv=None) -> int:
     params = Params.from_args(sys.argv[1:] if argv is None else argv)
+    shards = params.get_int("shards", 1)
+    if shards > 1:
+        import multiprocessing as mp
+        base = params.get_int("port", 6123)
+        ctx = mp.get_context("spawn")
+        procs = [ctx.Process(target=_run_shard,
+                             args=(params.to_dict(), base + s))
+                 for s in range(shards)]
+        for p in procs:
+            p.start()
+        print(f"serving {shards} key-partitioned shards on ports "
+              f"{base}..{base + shards - 1}")
+        for p in procs:
+            p.join()
+        return 0
     # fixed-delay restart strategy (consumer parity: 3 attempts / 10 s,
     # ALSKafkaConsumer.java:48-51); state survives restarts via the
     # checkpoint snapshots (restore by passing the snapshot as --alsModel)

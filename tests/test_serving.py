@@ -300,3 +300,59 @@ def test_attach_factors_lazy_payload():
     assert ok.tolist() == [True, False]
     exp = float(model.user_factors[1].double() @ model.item_factors[2].double())
     assert preds[0].item() == pytest.approx(exp, abs=0.05)  # bf16 mirror
+
+
+def test_sharded_serving(tmp_path):
+    """Key-partitioned serving across two live endpoints with client-side
+    routing (the reference's TaskManager state sharding)."""
+    import socket
+    import threading
+    import time as time_mod
+
+    import uvicorn
+
+    from flink_ms_amd.serving.sharding import ShardedQueryClient, shard_of
+
+    ports = []
+    servers = []
+    for _ in range(2):
+        s = socket.socket(); s.bind(("127.0.0.1", 0))
+        ports.append(s.getsockname()[1]); s.close()
+    for port in ports:
+        app = create_app(ALSModelStore(device=torch.device("cpu")),
+                         SVMModelStore())
+        srv = uvicorn.Server(uvicorn.Config(app, host="127.0.0.1",
+                                            port=port, log_level="error"))
+        threading.Thread(target=srv.run, daemon=True).start()
+        servers.append(srv)
+    client = ShardedQueryClient([("127.0.0.1", p) for p in ports])
+    for _ in range(100):
+        try:
+            client.clients[0]._client.get(
+                client.clients[0].base + "/healthz").raise_for_status()
+            client.clients[1]._client.get(
+                client.clients[1].base + "/healthz").raise_for_status()
+            break
+        except Exception:
+            time_mod.sleep(0.1)
+    rows = [f"{i},U,{float(i)};1.0" for i in range(20)]
+    rows += [f"{i},I,2.0;{float(i)}" for i in range(20)]
+    assert client.ingest_rows("als", rows) == 40
+    # every key lands on its hash shard and only there
+    for i in (0, 7, 13):
+        key = f"{i}-U"
+        home = shard_of(key, 2)
+        assert client.clients[home].query_state("ALS_MODEL", key) is not None
+        assert client.clients[1 - home].query_state("ALS_MODEL", key) is None
+        assert client.query_state("ALS_MODEL", key)[1].startswith(f"{i}.0")
+    # cross-shard predict: dot(U[3], V[5]) = 3*2 + 1*5
+    r = client.als_predict("3", "5")
+    assert r["found"] and r["prediction"] == pytest.approx(11.0)
+    assert not client.als_predict("3", "404")["found"]
+    # svm routing
+    assert client.ingest_rows("svm", ["1,0.5", "2,-1.0", "3,2.0"]) == 3
+    r = client.svm_predict("1:2.0 3:1.0", output_decision_function=True)
+    assert r["raw"] == pytest.approx(3.0) and not r["messages"]
+    client.close()
+    for srv in servers:
+        srv.should_exit = True
