@@ -92,6 +92,34 @@ def fixed_qps_als(port, qps, duration_s=5.0, workers=8,
             "p99_ms": pct(0.99)}
 
 
+def _qps_worker(ports, ids, iids, t0val, bar, qps, n, w, nproc, rq):
+    import random
+    import time as time_mod
+
+    from flink_ms_amd.serving.sharding import ShardedQueryClient
+
+    rng = random.Random(100 + w)
+    cl = ShardedQueryClient([("127.0.0.1", p) for p in ports], 5.0)
+    for c in cl.clients:  # open the connections before the clock starts
+        c._client.get(c.base + "/healthz")
+    bar.wait()   # all workers constructed
+    bar.wait()   # parent has published t0
+    t0 = t0val.value
+    lats = []
+    for k in range(w, n, nproc):
+        due = t0 + k / qps
+        now = time_mod.perf_counter()
+        if due > now:
+            time_mod.sleep(due - now)
+        s0 = time_mod.perf_counter()
+        r = cl.als_predict(rng.choice(ids), rng.choice(iids))
+        assert r["found"]
+        lats.append((time_mod.perf_counter() - s0) * 1000.0)
+    end = time_mod.perf_counter()
+    cl.close()
+    rq.put((lats, end))
+
+
 def sharded_qps(als_rows, n_shards=4, qps=4000, duration_s=5.0,
                 workers=16):
     """Key-partitioned serving: N shard processes + routed clients.
@@ -128,47 +156,41 @@ def sharded_qps(als_rows, n_shards=4, qps=4000, duration_s=5.0,
     probe.ingest_rows("als", als_rows)
     probe.close()
 
-    rng = random.Random(11)
+    # load from WORKER PROCESSES (a threaded client is GIL-bound and
+    # measures itself, not the servers).  perf_counter is CLOCK_MONOTONIC:
+    # system-wide, comparable across the workers.
     n = int(qps * duration_s)
-    t0 = time.perf_counter() + 0.3
-    jobs = queue_mod.Queue()
+    rq = ctx.Queue()
+    t0val = ctx.Value("d", 0.0)
+    nproc = 16
+    bar = ctx.Barrier(nproc + 1)
     ids = [r.split(",", 1)[0] for r in als_rows if ",U," in r]
     iids = [r.split(",", 1)[0] for r in als_rows if ",I," in r]
-    for k in range(n):
-        jobs.put((t0 + k / qps, rng.choice(ids), rng.choice(iids)))
-    lats = []
-    lock = th.Lock()
-
-    def worker():
-        cl = ShardedQueryClient([("127.0.0.1", p) for p in ports], 5.0)
-        while True:
-            try:
-                due, u, i = jobs.get_nowait()
-            except queue_mod.Empty:
-                cl.close()
-                return
-            now = time.perf_counter()
-            if due > now:
-                time.sleep(due - now)
-            s0 = time.perf_counter()
-            r = cl.als_predict(u, i)
-            assert r["found"]
-            with lock:
-                lats.append((time.perf_counter() - s0) * 1000.0)
-
-    threads = [th.Thread(target=worker) for _ in range(workers)]
-    start = time.perf_counter()
-    for t in threads:
-        t.start()
-    for t in threads:
-        t.join()
-    wall = time.perf_counter() - start
+    workers = [ctx.Process(target=_qps_worker,
+                           args=(ports, ids, iids, t0val, bar, qps, n, w,
+                                 nproc, rq),
+                           daemon=True)
+               for w in range(nproc)]
+    for w in workers:
+        w.start()
+    bar.wait()                                # every worker is constructed
+    t0val.value = time.perf_counter() + 0.2   # common schedule origin
+    bar.wait()                                # release the open loop
+    lats, end = [], 0.0
+    for _ in range(nproc):
+        wl, wend = rq.get(timeout=duration_s * 10 + 120)
+        lats.extend(wl)
+        end = max(end, wend)
+    for w in workers:
+        w.join(timeout=30)
+    wall = end - t0val.value
     for p in procs:
         p.terminate()
     lats.sort()
 
-    def pct(x):
-        return lats[min(len(lats) - 1, int(x * len(lats)))] if lats else None
+    def pct(q):
+        return lats[min(len(lats) - 1, int(q * len(lats)))]
+
     return {"shards": n_shards, "target_qps": qps,
             "achieved_qps": len(lats) / wall, "p50_ms": pct(0.5),
             "p95_ms": pct(0.95)}
@@ -218,7 +240,7 @@ def main():
 
     # key-partitioned scale-out (4 shard processes, routed clients)
     rows = als.snapshot_rows()
-    results["als_http_sharded4"] = sharded_qps(rows, n_shards=4, qps=6000)
+    results["als_http_sharded4"] = sharded_qps(rows, n_shards=4, qps=8000)
     print(json.dumps(results, indent=1), flush=True)
 
 
