@@ -101,8 +101,14 @@ def als_solve_side(
             # 1B-rating configs would exceed HBM if materialized whole.
             # A sliced indptr keeps GLOBAL offsets into indices/values, so
             # each slab call reuses the full nnz arrays untouched.
-            slab = slab_rows or max(1, min(csr.num_rows,
-                                           (1 << 30) // (k * k * 4)))
+            # Measured (gpu_debug/slab_pipeline_timing.py): bigger slabs
+            # are monotonically faster (fewer launch/tail bubbles), so the
+            # cap is generous — 4 GiB of A in 288 GB HBM3E; two-stream
+            # gramian/solve pipelining across slabs was NEUTRAL at every
+            # slab size (the gramian saturates the CUs on its own) and is
+            # kept out.
+            cap = max(1, (4 << 30) // (k * k * 4))
+            slab = slab_rows or max(1, min(csr.num_rows, cap))
             out = torch.empty(csr.num_rows, k, dtype=torch.float32,
                               device=fac.device)
             A = torch.empty(slab, k, k, dtype=torch.float32,
