@@ -238,16 +238,38 @@ __global__ void k_ldl_solve_wave(const float* __restrict__ A_in,
                                  long long nrows) {
     constexpr int K = KT * 16;
     static_assert(K <= 64, "wave solver handles k <= 64");
-    constexpr int LDA = K + 1;
-    __shared__ __align__(16) float As[4][K * LDA];
+    // A lives in LDS as LOWER-TRIANGLE 16x16 tiles only (tile (I,J), I>=J,
+    // at triangular offset I(I+1)/2+J).  The LDL never reads above the
+    // block diagonal; the panel factorization's upper-scratch trick only
+    // needs the diagonal tiles' upper halves, which the full 16x16 tiles
+    // keep.  At K=64 this is 10/16 of the square: LDS 66.5 -> 46 KB per
+    // block (2 -> 3 waves/SIMD) and 10/16 of the HBM read of A.
+    // Tile row stride 18: 8-byte-aligned rows, conflict-free column access
+    // (18*i mod 32 distinct over i=0..15).
+    constexpr int NP = K / 16;
+    constexpr int NTRI = NP * (NP + 1) / 2;
+    constexpr int LDT = 18;
+    constexpr int TSZ = 16 * LDT;
+    __shared__ __align__(16) float As[4][NTRI * TSZ];
     const int w = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
     const int lane = threadIdx.x & 63;
     const long long e = (long long)blockIdx.x * 4 + w;
     if (e >= nrows) return;
     float* A = As[w];
-    const float* src = A_in + e * (long long)(K * K);
-    for (int i = lane; i < K * K; i += 64)
-        A[(i / K) * LDA + (i % K)] = src[i];
+    const int li = lane & 15, g4 = lane >> 4;
+    {   // load the lower tiles: one whole tile per iteration (16 rows x
+        // 4 float4 quads = 64 lanes), contiguous 16-byte global reads
+        const float* src = A_in + e * (long long)(K * K);
+        const int r = lane >> 2, c4 = lane & 3;
+        int I = 0, J = 0;
+        for (int t = 0; t < NTRI; ++t) {
+            const f32x4 v =
+                *(const f32x4*)(src + (I * 16 + r) * K + J * 16 + c4 * 4);
+            float* dst = A + t * TSZ + r * LDT + c4 * 4;
+            dst[0] = v[0]; dst[1] = v[1]; dst[2] = v[2]; dst[3] = v[3];
+            if (++J > I) { ++I; J = 0; }
+        }
+    }
     float x0 = (lane < K) ? b_in[e * K + lane] : 0.0f;
     // In-place LDL^T elimination, 16-column panels with MFMA trailing
     // updates (cdna_hip_programming.md §6 G10: the trailing update
@@ -267,20 +289,24 @@ __global__ void k_ldl_solve_wave(const float* __restrict__ A_in,
     // panel factorization is masked to lane < K; every address in the
     // MFMA phase is bounded by K for all lanes.
     {
-        constexpr int NP = K / 16;
-        float* rowp = A + lane * LDA;   // only dereferenced when lane < K
-        const int li = lane & 15, g4 = lane >> 4;
+        auto tri = [](int I, int J) { return (I * (I + 1)) / 2 + J; };
+        // rowseg: lane = global row; its 16-col panel-pi segment lives in
+        // tile (lane>>4, pi).  Only dereferenced when P0 <= lane < K.
         for (int pi = 0; pi < NP; ++pi) {
             const int P0 = 16 * pi;
             // (a) panel factorization, fully in registers: each lane holds
             // its row's 16-column panel segment; the pivot column A[c][j]
             // lives in lane c's seg[jj], broadcast by __shfl -- the inner
-            // rank-1 update is shfl+fma with ZERO LDS traffic.  (For the
-            // last panel j runs to P0+14 = K-2, same bound.)
-            if (lane < K) {
+            // rank-1 update is shfl+fma with ZERO LDS traffic.  Rows above
+            // the panel (lane < P0) have no storage in the triangular
+            // image and sit the phase out (one exec mask per panel, not
+            // per element).
+            if (lane >= P0 && lane < K) {
+                float* rowseg =
+                    A + tri(lane >> 4, pi) * TSZ + (lane & 15) * LDT;
                 float seg[16];
 #pragma unroll
-                for (int cc = 0; cc < 16; ++cc) seg[cc] = rowp[P0 + cc];
+                for (int cc = 0; cc < 16; ++cc) seg[cc] = rowseg[cc];
 #pragma unroll
                 for (int jj = 0; jj < 15; ++jj) {
                     const float dj = __shfl(seg[jj], P0 + jj, WAVE);
@@ -293,50 +319,62 @@ __global__ void k_ldl_solve_wave(const float* __restrict__ A_in,
                     }
                 }
 #pragma unroll
-                for (int cc = 0; cc < 16; ++cc) rowp[P0 + cc] = seg[cc];
+                for (int cc = 0; cc < 16; ++cc) rowseg[cc] = seg[cc];
             }
             if (pi == NP - 1) break;
             // (b) MFMA trailing update, whole wave.  -1/D per contraction
-            // column, col(kk) = P0 + 4*kk + g4.
+            // column, local panel col 4*kk + g4.
+            const float* Tpp = A + tri(pi, pi) * TSZ;
             float ndk[4];
 #pragma unroll
             for (int kk = 0; kk < 4; ++kk) {
-                const int c = P0 + 4 * kk + g4;
-                const float d = A[c * LDA + c];
+                const int c = 4 * kk + g4;
+                const float d = Tpp[c * LDT + c];
                 ndk[kk] = d > 0.0f ? -1.0f / d : 0.0f;
             }
             for (int rb = pi + 1; rb < NP; ++rb) {
+                const float* TA = A + tri(rb, pi) * TSZ;  // L21 rows rb
                 for (int cb = pi + 1; cb <= rb; ++cb) {
+                    const float* TB = A + tri(cb, pi) * TSZ;
+                    float* TC = A + tri(rb, cb) * TSZ;
                     f32x4 acc;   // C tile: D map row=(l>>4)*4+r, col=l&15
 #pragma unroll
                     for (int r = 0; r < 4; ++r)
-                        acc[r] = A[(rb * 16 + g4 * 4 + r) * LDA + cb * 16 + li];
+                        acc[r] = TC[(g4 * 4 + r) * LDT + li];
 #pragma unroll
                     for (int kk = 0; kk < 4; ++kk) {
-                        const int pc = P0 + 4 * kk + g4;
-                        const float a = A[(rb * 16 + li) * LDA + pc];
-                        const float b = A[(cb * 16 + li) * LDA + pc] * ndk[kk];
+                        const int pc = 4 * kk + g4;
+                        const float a = TA[li * LDT + pc];
+                        const float b = TB[li * LDT + pc] * ndk[kk];
                         acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc,
                                                                    0, 0, 0);
                     }
 #pragma unroll
                     for (int r = 0; r < 4; ++r)
-                        A[(rb * 16 + g4 * 4 + r) * LDA + cb * 16 + li] = acc[r];
+                        TC[(g4 * 4 + r) * LDT + li] = acc[r];
                 }
             }
         }
     }
-    // solve (I+Ls) y = b ; w = y/D ; (I+Ls^T) x = w, all in registers
-    const float d0 = (lane < K) ? A[lane * LDA + lane] : 1.0f;
+    // solve (I+Ls) y = b ; w = y/D ; (I+Ls^T) x = w, all in registers.
+    // Row `lane`'s tiles sit at triangular offsets rowbase + J for J <=
+    // lane>>4 (rowbase fixed per lane).
+    const int Irow = lane >> 4;
+    const float* rowbase = A + ((Irow * (Irow + 1)) / 2) * TSZ + li * LDT;
+    const float d0 = (lane < K) ? rowbase[Irow * TSZ + li] : 1.0f;
     const float id0 = d0 > 0.0f ? 1.0f / d0 : 0.0f;
     for (int j = 0; j < K - 1; ++j) {
         const float zj = __shfl(x0, j, WAVE) * __shfl(id0, j, WAVE);
-        if (lane > j && lane < K) x0 -= A[lane * LDA + j] * zj;
+        if (lane > j && lane < K)
+            x0 -= rowbase[(j >> 4) * TSZ + (j & 15)] * zj;
     }
     x0 *= id0;
     for (int c = K - 1; c >= 1; --c) {
         const float xc = __shfl(x0, c, WAVE);
-        if (lane < c) x0 -= A[c * LDA + lane] * id0 * xc;
+        // A(c, lane): row c's tile at column block lane>>4
+        if (lane < c)
+            x0 -= A[(((c >> 4) * ((c >> 4) + 1)) / 2 + (lane >> 4)) * TSZ
+                    + (c & 15) * LDT + li] * id0 * xc;
     }
     if (lane < K) {
         x_out[e * K + lane] = x0;
