@@ -13,6 +13,7 @@ TaskManagers); clients and producers route by
 ``serving.sharding.shard_of`` (ShardedQueryClient).
 """
 import sys
+import time
 
 import uvicorn
 
@@ -67,7 +68,6 @@ def main(argv=None) -> int:
     # checkpoint snapshots (restore by passing the snapshot as --alsModel)
     attempts = params.get_int("restartAttempts", 3)
     delay_s = params.get_int("restartDelay", 10000) / 1000.0
-    import time
     for attempt in range(attempts + 1):
         try:
             app = build_app(params)

@@ -106,6 +106,23 @@ def create_app(als_store: Optional[ALSModelStore] = None,
     stop_evt = threading.Event()
     app.state._ckpt_stop = stop_evt
 
+    # Checkpoint RESTORE (Flink parity: the serving job restores its keyed
+    # state from the latest completed checkpoint on restart,
+    # ALSKafkaConsumer.java:44-51 enableCheckpointing + restart strategy).
+    # Snapshots are model-text rows, so restore is an ingest of the newest
+    # per-store snapshot.  Only empty stores restore — explicitly preloaded
+    # models (--alsModel/--svmModel) win.
+    if checkpoint_data_uri and os.path.isdir(checkpoint_data_uri):
+        for name, store in (("als", als), ("svm", svm)):
+            if len(store):
+                continue
+            snaps = sorted(f for f in os.listdir(checkpoint_data_uri)
+                           if f.startswith(name + "-")
+                           and f.endswith(".model"))
+            if snaps:
+                with open(os.path.join(checkpoint_data_uri, snaps[-1])) as f:
+                    store.ingest([ln for ln in f.read().splitlines() if ln])
+
     def _checkpoint() -> dict:
         if not app.state.checkpoint_uri:
             return {"written": 0}

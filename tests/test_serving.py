@@ -2,6 +2,7 @@
 online SGD loop, load generators, checkpointing."""
 
 import io
+import time
 
 import pytest
 import torch
@@ -356,3 +357,54 @@ def test_sharded_serving(tmp_path):
     client.close()
     for srv in servers:
         srv.should_exit = True
+
+
+def test_checkpoint_auto_restore_on_restart(tmp_path, als_store, svm_store):
+    """Flink parity: a restarted serving job restores keyed state from the
+    latest completed checkpoint (enableCheckpointing + fixed-delay restart,
+    ALSKafkaConsumer.java:44-51)."""
+    uri = str(tmp_path / "ckpt")
+    app = create_app(als_store, svm_store, checkpoint_data_uri=uri,
+                     checkpoint_interval_ms=0)
+    c = TestClient(app)
+    c.post("/model/als/rows", json={"rows": ["77,U,9.0;9.0;9.0"]})
+    assert c.post("/checkpoint").json()["written"] > 0
+    time.sleep(0.002)  # distinct stamp for the second checkpoint
+    c.post("/model/als/rows", json={"rows": ["77,U,1.0;2.0;3.0"]})
+    c.post("/checkpoint")
+
+    # "restart": fresh empty stores, same checkpointDataUri -> the NEWEST
+    # snapshot restores
+    app2 = create_app(checkpoint_data_uri=uri, checkpoint_interval_ms=0)
+    c2 = TestClient(app2)
+    assert c2.get("/state/ALS_MODEL/77-U").json()["value"][1] == "1.0;2.0;3.0"
+    assert c2.get("/state/ALS_MODEL/1-U").json() == \
+        c.get("/state/ALS_MODEL/1-U").json()
+    assert c2.get("/state/SVM_MODEL/2").status_code == 200
+
+    # explicitly preloaded stores WIN over snapshots (--alsModel parity)
+    pre = ALSModelStore(device=torch.device("cpu"))
+    pre.ingest(["1,U,5.0;5.0;5.0"])
+    app3 = create_app(pre, checkpoint_data_uri=uri, checkpoint_interval_ms=0)
+    assert TestClient(app3).get("/state/ALS_MODEL/1-U").json()["value"][1] \
+        == "5.0;5.0;5.0"
+
+
+def test_serve_cli_fixed_delay_restart(monkeypatch, capsys):
+    """cli/serve retries uvicorn.run failures restartAttempts times with the
+    fixed delay (consumer restart-strategy parity)."""
+    from flink_ms_amd.cli import serve as serve_cli
+
+    calls = {"n": 0}
+
+    def flaky_run(app, **kw):
+        calls["n"] += 1
+        if calls["n"] < 3:
+            raise RuntimeError("bind failed")
+
+    monkeypatch.setattr(serve_cli.uvicorn, "run", flaky_run)
+    monkeypatch.setattr(serve_cli.time, "sleep", lambda s: None)
+    assert serve_cli.main(["--restartAttempts", "3",
+                           "--restartDelay", "10"]) == 0
+    assert calls["n"] == 3
+    assert "restart 1/3" in capsys.readouterr().out
