@@ -52,20 +52,20 @@ __global__ void k_als_solve_fused(const long long* __restrict__ indptr,
                                   const int* __restrict__ row_order,
                                   long long nrows, float reg) {
     constexpr int K = Geo<KT>::K;
-    __shared__ __align__(16) char smem[Geo<KT>::SMEM];
+    __shared__ __align__(16) char smem[Geo<KT>::SMEM_TRI];
     long long row = blockIdx.x;
     if (row >= nrows) return;
     if (row_order) row = row_order[row];
 
-    const int n = gramian_to_lds<KT>(smem, indptr, indices, values, factors,
-                                     row, reg);
+    const int n = gramian_to_lds<KT, true, true>(smem, indptr, indices,
+                                                 values, factors, row, reg);
     if (n == 0) { write_zero_row<KT>(out_f32, out_bf16, row); return; }
 
     float* A = (float*)smem;
-    float* b = A + K * (K + 1);
-    cholesky_lds<K>(A, b + 2 * K);
-    solve_lds_block<K>(A, b, b + 2 * K, out_f32 + row * K,
-                       out_bf16 ? out_bf16 + row * K : nullptr);
+    float* b = A + Geo<KT>::A_TRI_FLOATS;
+    cholesky_lds_tri<K>(A, b + 2 * K);
+    solve_lds_block_tri<K>(A, b, b + 2 * K, out_f32 + row * K,
+                           out_bf16 ? out_bf16 + row * K : nullptr);
 }
 
 // Standalone K1 (for parity tests / modular path): writes dense A and b.
@@ -107,19 +107,29 @@ __global__ void k_cholesky_solve(const float* __restrict__ A_in,  // [n][K][K]
                                  float* __restrict__ x_out,       // [n][K]
                                  long long nrows, int phases) {
     constexpr int K = Geo<KT>::K;
-    __shared__ __align__(16) char smem[Geo<KT>::SMEM];
+    constexpr int NTRI = Geo<KT>::NA;
+    __shared__ __align__(16) char smem[Geo<KT>::CHOL_TRI_BYTES];
     const long long row = blockIdx.x;
     if (row >= nrows) return;
     float* A = (float*)smem;
-    float* b = A + K * (K + 1);
+    float* b = A + Geo<KT>::A_TRI_FLOATS;
     const int tid = threadIdx.x;
-    for (int i = tid; i < K * K; i += 256)
-        A[(i / K) * (K + 1) + (i % K)] = A_in[row * K * K + i];
+    {   // load the lower tiles of the (symmetric) square A into the tri
+        // image: one whole 16x16 tile per iteration of the 256 threads
+        const float* src = A_in + row * (long long)(K * K);
+        const int r = tid >> 4, c = tid & 15;
+        int I = 0, J = 0;
+        for (int t = 0; t < NTRI; ++t) {
+            A[t * Geo<KT>::TSZ + r * Geo<KT>::LDT + c] =
+                src[(I * 16 + r) * K + J * 16 + c];
+            if (++J > I) { ++I; J = 0; }
+        }
+    }
     for (int c = tid; c < K; c += 256) b[c] = b_in[row * K + c];
     __syncthreads();
-    if (phases & 1) cholesky_lds<K>(A, b + 2 * K);
+    if (phases & 1) cholesky_lds_tri<K>(A, b + 2 * K);
     if (phases & 2) {
-        solve_lds_block<K>(A, b, b + 2 * K, x_out + row * K, nullptr);
+        solve_lds_block_tri<K>(A, b, b + 2 * K, x_out + row * K, nullptr);
     } else if (tid < K) {
         x_out[row * K + tid] = b[tid];
     }
