@@ -125,14 +125,10 @@ def sharded_qps(als_rows, n_shards=4, qps=4000, duration_s=5.0,
     """Key-partitioned serving: N shard processes + routed clients.
     Measures aggregate sustained QPS + latency at a fixed offered rate."""
     import multiprocessing as mp
-    import queue as queue_mod
-    import random
-    import threading as th
 
     from flink_ms_amd.cli.serve import _run_shard
     from flink_ms_amd.serving.sharding import ShardedQueryClient
 
-    base = None
     socks = []
     ports = []
     for _ in range(n_shards):
@@ -162,7 +158,7 @@ def sharded_qps(als_rows, n_shards=4, qps=4000, duration_s=5.0,
     n = int(qps * duration_s)
     rq = ctx.Queue()
     t0val = ctx.Value("d", 0.0)
-    nproc = 16
+    nproc = workers
     bar = ctx.Barrier(nproc + 1)
     ids = [r.split(",", 1)[0] for r in als_rows if ",U," in r]
     iids = [r.split(",", 1)[0] for r in als_rows if ",I," in r]
