@@ -236,7 +236,10 @@ def main():
 
     # key-partitioned scale-out (4 shard processes, routed clients)
     rows = als.snapshot_rows()
+    results["als_http_sharded2"] = sharded_qps(rows, n_shards=2, qps=4000)
     results["als_http_sharded4"] = sharded_qps(rows, n_shards=4, qps=8000)
+    results["als_http_sharded8"] = sharded_qps(rows, n_shards=8, qps=16000,
+                                               workers=24)
     print(json.dumps(results, indent=1), flush=True)
 
 
