@@ -121,7 +121,7 @@ def als_solve_side(
                             fac, As, bs, float(reg), _stream())
                 obs = ob[s:e] if ob.numel() > 0 else ob
                 if k <= 64:
-                    ops.ldl_solve_wave(As, bs, out[s:e], obs, _stream())
+                    ops.ldl_solve_wave_reg(As, bs, out[s:e], obs, _stream())
                 else:
                     ops.cholesky_solve(As, bs, out[s:e], _stream())
                     if out_bf16 is not None:

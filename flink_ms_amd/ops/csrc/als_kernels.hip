@@ -392,6 +392,229 @@ __global__ void k_ldl_solve_wave(const float* __restrict__ A_in,
     }
 }
 
+
+// -------- register-resident wave LDL (k <= 64), v2 --------
+// A lives in REGISTERS as the 10 lower-triangle 16x16 MFMA C-fragments
+// (lane holds (row = (l>>4)*4 + r, col = l&15) of each tile); a 4.35 KB
+// per-entity LDS scratch re-shapes one 16-column panel (or one tile
+// row/column for the substitution) at a time.  LDS per 4-entity block
+// drops 46 -> 17.4 KB and the trailing-update C tiles never touch LDS.
+// All tile indices are COMPILE-TIME (constexpr recursion): runtime
+// indexing of the fragment array would spill to scratch (guide rule 20).
+// Scratch strides: 17 (panel rows, conflict-free columns), 65 (tile-row
+// image for the backward substitution).
+// WREG_FENCE: the scratch bounces are cross-lane LDS write->read chains
+// with no barrier (single wave).  The LDS unit processes a wave's ds ops
+// in order, but the COMPILER must not reorder the reads above the masked
+// writes (observed miscompile: reader lanes in other row-quads got stale
+// panel values) -- a compiler memory fence plus lgkmcnt drain pins the
+// order at negligible cost (a handful per entity).
+#define WREG_FENCE() __asm__ volatile("s_waitcnt lgkmcnt(0)" ::: "memory")
+
+
+template <int KT, int PI, int I>
+DEV_INLINE void wreg_dump_col(const f32x4* T, float* scr, int g4, int li) {
+    if constexpr (I < KT) {
+        constexpr int t = tri_off(I, PI);
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+            scr[((I - PI) * 16 + g4 * 4 + r) * 17 + li] = T[t][r];
+        wreg_dump_col<KT, PI, I + 1>(T, scr, g4, li);
+    }
+}
+
+template <int KT, int PI, int I>
+DEV_INLINE void wreg_load_col(f32x4* T, const float* scr, int g4, int li) {
+    if constexpr (I < KT) {
+        constexpr int t = tri_off(I, PI);
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+            T[t][r] = scr[((I - PI) * 16 + g4 * 4 + r) * 17 + li];
+        wreg_load_col<KT, PI, I + 1>(T, scr, g4, li);
+    }
+}
+
+template <int KT, int PI, int RB, int CB>
+DEV_INLINE void wreg_trail(f32x4* T, const float* scr, const float* ndk,
+                           int g4, int li) {
+    if constexpr (RB < KT) {
+        if constexpr (CB > RB) {
+            wreg_trail<KT, PI, RB + 1, PI + 1>(T, scr, ndk, g4, li);
+        } else {
+            constexpr int t = tri_off(RB, CB);
+            f32x4 acc = T[t];
+#pragma unroll
+            for (int kk = 0; kk < 4; ++kk) {
+                const int pc = 4 * kk + g4;
+                const float a = scr[((RB - PI) * 16 + li) * 17 + pc];
+                const float b =
+                    scr[((CB - PI) * 16 + li) * 17 + pc] * ndk[kk];
+                acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0,
+                                                           0);
+            }
+            T[t] = acc;
+            wreg_trail<KT, PI, RB, CB + 1>(T, scr, ndk, g4, li);
+        }
+    }
+}
+
+template <int KT, int PI>
+DEV_INLINE void wreg_panels(f32x4* T, float* scr, int lane, int g4, int li,
+                            float& d0) {
+    if constexpr (PI < KT) {
+        constexpr int K = KT * 16;
+        constexpr int P0 = 16 * PI;
+        wreg_dump_col<KT, PI, PI>(T, scr, g4, li);
+        WREG_FENCE();
+        // register factorization of the panel: lane = row P0+lane
+        if (lane < K - P0) {
+            float seg[16];
+#pragma unroll
+            for (int cc = 0; cc < 16; ++cc) seg[cc] = scr[lane * 17 + cc];
+#pragma unroll
+            for (int jj = 0; jj < 15; ++jj) {
+                const float dj = __shfl(seg[jj], jj, WAVE);
+                const float dinv = dj > 0.0f ? 1.0f / dj : 0.0f;
+                const float lscl = seg[jj] * dinv;
+#pragma unroll
+                for (int cc = jj + 1; cc < 16; ++cc) {
+                    const float colj = __shfl(seg[jj], cc, WAVE);
+                    seg[cc] -= lscl * colj;
+                }
+            }
+#pragma unroll
+            for (int cc = 0; cc < 16; ++cc) scr[lane * 17 + cc] = seg[cc];
+        }
+        WREG_FENCE();
+        if (lane >= P0 && lane < P0 + 16)
+            d0 = scr[(lane - P0) * 17 + (lane - P0)];
+        wreg_load_col<KT, PI, PI>(T, scr, g4, li);
+        if constexpr (PI + 1 < KT) {
+            float ndk[4];
+#pragma unroll
+            for (int kk = 0; kk < 4; ++kk) {
+                const int c = 4 * kk + g4;
+                const float d = scr[c * 17 + c];
+                ndk[kk] = d > 0.0f ? -1.0f / d : 0.0f;
+            }
+            wreg_trail<KT, PI, PI + 1, PI + 1>(T, scr, ndk, g4, li);
+        }
+        wreg_panels<KT, PI + 1>(T, scr, lane, g4, li, d0);
+    }
+}
+
+template <int KT, int J>
+DEV_INLINE void wreg_forward(const f32x4* T, float* scr, float& x0,
+                             float id0, int lane, int g4, int li) {
+    if constexpr (J < KT) {
+        constexpr int K = KT * 16;
+        constexpr int B0 = 16 * J;
+        wreg_dump_col<KT, J, J>(T, scr, g4, li);
+        WREG_FENCE();
+#pragma unroll
+        for (int t = 0; t < 16; ++t) {
+            const int j = B0 + t;
+            if (j >= K - 1) break;
+            const float zj = __shfl(x0, j, WAVE) * __shfl(id0, j, WAVE);
+            if (lane > j && lane < K)
+                x0 -= scr[(lane - B0) * 17 + t] * zj;
+        }
+        wreg_forward<KT, J + 1>(T, scr, x0, id0, lane, g4, li);
+    }
+}
+
+template <int KT, int I, int J>
+DEV_INLINE void wreg_dump_row(const f32x4* T, float* scr, int g4, int li) {
+    if constexpr (J <= I) {
+        constexpr int t = tri_off(I, J);
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+            scr[(g4 * 4 + r) * 65 + J * 16 + li] = T[t][r];
+        wreg_dump_row<KT, I, J + 1>(T, scr, g4, li);
+    }
+}
+
+template <int KT, int I>
+DEV_INLINE void wreg_backward(const f32x4* T, float* scr, float& x0,
+                              float id0, int lane, int g4, int li) {
+    if constexpr (I >= 0) {
+        constexpr int B0 = 16 * I;
+        wreg_dump_row<KT, I, 0>(T, scr, g4, li);
+        WREG_FENCE();
+#pragma unroll
+        for (int t = 15; t >= 0; --t) {
+            const int c = B0 + t;
+            if (c < 1) break;
+            const float xc = __shfl(x0, c, WAVE);
+            if (lane < c)
+                x0 -= scr[t * 65 + lane] * id0 * xc;
+        }
+        wreg_backward<KT, I - 1>(T, scr, x0, id0, lane, g4, li);
+    }
+}
+
+template <int KT, int t>
+DEV_INLINE void wreg_load_A(f32x4* T, const float* src, int g4, int li) {
+    if constexpr (t < KT * (KT + 1) / 2) {
+        constexpr int K = KT * 16;
+        constexpr int I = lo_tile_i(t);
+        constexpr int J = t - (I * (I + 1)) / 2;
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+            T[t][r] = src[(I * 16 + g4 * 4 + r) * K + J * 16 + li];
+        wreg_load_A<KT, t + 1>(T, src, g4, li);
+    }
+}
+
+template <int KT>
+__launch_bounds__(256)
+__global__ void k_ldl_solve_wave_reg(const float* __restrict__ A_in,
+                                     const float* __restrict__ b_in,
+                                     float* __restrict__ x_out,
+                                     unsigned short* __restrict__ x_bf16,
+                                     long long nrows) {
+    constexpr int K = KT * 16;
+    static_assert(K <= 64, "register wave solver handles k <= 64");
+    constexpr int NTRI = KT * (KT + 1) / 2;
+    constexpr int SCR = K * 17;            // covers 16*65=1040 too (K=64)
+    __shared__ __align__(16) float As[4][SCR > 1040 ? SCR : 1040];
+    const int w = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
+    const int lane = threadIdx.x & 63;
+    const long long e = (long long)blockIdx.x * 4 + w;
+    if (e >= nrows) return;
+    float* scr = As[w];
+    const int li = lane & 15, g4 = lane >> 4;
+
+    f32x4 T[NTRI];
+    wreg_load_A<KT, 0>(T, A_in + e * (long long)(K * K), g4, li);
+    float x0 = (lane < K) ? b_in[e * K + lane] : 0.0f;
+    float d0 = 1.0f;
+    wreg_panels<KT, 0>(T, scr, lane, g4, li, d0);
+    const float id0 = d0 > 0.0f ? 1.0f / d0 : 0.0f;
+    wreg_forward<KT, 0>(T, scr, x0, id0, lane, g4, li);
+    x0 *= id0;
+    wreg_backward<KT, KT - 1>(T, scr, x0, id0, lane, g4, li);
+    if (lane < K) {
+        x_out[e * K + lane] = x0;
+        if (x_bf16) x_bf16[e * K + lane] = f2bf(x0);
+    }
+}
+
+extern "C" hipError_t fma_ldl_solve_wave_reg(
+    int k, const float* A_in, const float* b_in, float* x_out,
+    unsigned short* x_bf16, long long nrows, hipStream_t stream) {
+    if (k % 16 || k < 16 || k > 64 || nrows <= 0) return hipErrorInvalidValue;
+    dim3 grid((unsigned)((nrows + 3) / 4)), block(256);
+    switch (k / 16) {
+        case 1: k_ldl_solve_wave_reg<1><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, nrows); break;
+        case 2: k_ldl_solve_wave_reg<2><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, nrows); break;
+        case 3: k_ldl_solve_wave_reg<3><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, nrows); break;
+        case 4: k_ldl_solve_wave_reg<4><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, nrows); break;
+        default: return hipErrorInvalidValue;
+    }
+    return hipGetLastError();
+}
+
 extern "C" hipError_t fma_ldl_solve_wave(
     int k, const float* A_in, const float* b_in, float* x_out,
     unsigned short* x_bf16, long long nrows, hipStream_t stream) {

@@ -42,6 +42,9 @@ int fma_sgd_update(unsigned short* U, unsigned short* V,
 int fma_ldl_solve_wave(int k, const float* A_in, const float* b_in,
                        float* x_out, unsigned short* x_bf16, int64_t nrows,
                        void* stream);
+int fma_ldl_solve_wave_reg(int k, const float* A_in, const float* b_in,
+                           float* x_out, unsigned short* x_bf16,
+                           int64_t nrows, void* stream);
 int fma_mfma_probe_f32(const float* A, const float* B, float* D, void* stream);
 int fma_mfma_probe_bf16(const unsigned short* Xt, const unsigned short* Yt,
                         float* C, void* stream);
@@ -144,6 +147,16 @@ void ldl_solve_wave(torch::Tensor A, torch::Tensor b, torch::Tensor x,
                                  b.data_ptr<float>(), x.data_ptr<float>(),
                                  xb, A.size(0), (void*)stream),
               "ldl_solve_wave");
+}
+
+// register-resident v2 (A in MFMA fragments, panel scratch in LDS)
+void ldl_solve_wave_reg(torch::Tensor A, torch::Tensor b, torch::Tensor x,
+                        torch::Tensor x_bf16, int64_t stream) {
+    unsigned short* xb = x_bf16.numel() > 0 ? bf16_ptr_mut(x_bf16) : nullptr;
+    check_hip(fma_ldl_solve_wave_reg((int)A.size(1), A.data_ptr<float>(),
+                                     b.data_ptr<float>(), x.data_ptr<float>(),
+                                     xb, A.size(0), (void*)stream),
+              "ldl_solve_wave_reg");
 }
 
 void cholesky_solve(torch::Tensor A, torch::Tensor b, torch::Tensor x,
@@ -300,6 +313,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("cholesky_solve", &cholesky_solve);
     m.def("cholesky_solve_ph", &cholesky_solve_ph);
     m.def("ldl_solve_wave", &ldl_solve_wave);
+    m.def("ldl_solve_wave_reg", &ldl_solve_wave_reg);
     m.def("sdca_pass", &sdca_pass);
     m.def("svm_margins", &svm_margins);
     m.def("predict_dot", &predict_dot);
