@@ -234,3 +234,31 @@ def test_distributed_cli_training(tmp_path):
     assert als_mean_vector.main(["--type", "user", "--input", str(uf),
                                  "--output", str(out)]) == 0
     assert open(out).read().startswith("MEAN,U,")
+
+
+@pytest.mark.timeout(300)
+def test_driver_bench_launch_contract(tmp_path):
+    """The benchmark driver's exact multi-rank launch: torch.distributed.run
+    of bench.py at world_size=2 (gloo/CPU here; RCCL on the GPU node).
+    Rank 0 must print ONE JSON line with the whole-job aggregate."""
+    import json
+    import subprocess
+    import sys
+
+    res = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29779", "bench.py", "--gpus", "2",
+         "--steps", "2", "--warmup", "1", "--device", "cpu",
+         "--users-per-gpu", "300", "--items", "200",
+         "--ratings-per-gpu", "5000", "--rank", "16"],
+        capture_output=True, text=True, timeout=240)
+    assert res.returncode == 0, res.stderr[-2000:]
+    json_lines = [ln for ln in res.stdout.splitlines()
+                  if ln.startswith("{")]
+    assert len(json_lines) == 1, res.stdout
+    d = json.loads(json_lines[0])
+    assert d["n_gpus"] == 2 and d["steps"] == 2 and d["scaling"] == "weak"
+    assert d["config"]["global_batch"] == 10_000  # whole-job aggregate
+    assert d["value"] > 0 and d["ms_per_step"] > 0
+    assert d["config"]["parallelism"] == "dp2+factor-allgather"
