@@ -10,7 +10,9 @@ memory + model-format snapshots), --checkPointInterval (60000), plus
 ``--shards N`` key-partitions the state across N serving processes on
 ports port..port+N-1 (the reference shards keyed state across
 TaskManagers); clients and producers route by
-``serving.sharding.shard_of`` (ShardedQueryClient).
+``serving.sharding.shard_of`` (ShardedQueryClient).  ``--device`` pins
+the store's device mirror; with ``--spreadShards`` shard s uses
+``cuda:(s mod num_gpus)`` so the N processes spread over the node's GPUs.
 """
 import sys
 import time
@@ -23,8 +25,13 @@ from ..utils.params import Params
 
 
 def build_app(params: Params):
-    als = ALSModelStore()
-    svm = SVMModelStore()
+    import torch
+
+    dev = None
+    if params.has("device"):
+        dev = torch.device(params.get("device"))
+    als = ALSModelStore(device=dev)
+    svm = SVMModelStore(device=dev)
     if params.has("alsModel"):
         als.ingest(_read_rows(params.get("alsModel")))
     if params.has("svmModel"):
@@ -53,9 +60,17 @@ def main(argv=None) -> int:
         import multiprocessing as mp
         base = params.get_int("port", 6123)
         ctx = mp.get_context("spawn")
-        procs = [ctx.Process(target=_run_shard,
-                             args=(params.to_dict(), base + s))
-                 for s in range(shards)]
+        spread = params.get_bool("spreadShards", False)
+        procs = []
+        for s in range(shards):
+            pd = params.to_dict()
+            if spread and not params.has("device"):
+                import torch
+                n = torch.cuda.device_count()
+                if n > 0:
+                    pd["device"] = f"cuda:{s % n}"
+            procs.append(ctx.Process(target=_run_shard,
+                                     args=(pd, base + s)))
         for p in procs:
             p.start()
         print(f"serving {shards} key-partitioned shards on ports "

@@ -309,7 +309,9 @@ class SVMModelStore:
     """Keyed store: flat ``"<featureIdx>" -> weight`` or range-partitioned
     ``"<bucket>" -> "i:w;i:w;..."`` (SVMKafkaConsumer.java:74-92)."""
 
-    def __init__(self):
+    def __init__(self, device: Optional[torch.device] = None):
+        # device accepted for store-construction symmetry; the SVM state is
+        # scalar text weights served from host dicts
         self._payload: Dict[str, str] = {}
         self._flat: Dict[str, float] = {}                 # id -> weight
         self._buckets: Dict[str, Dict[str, float]] = {}   # bucket -> id -> w

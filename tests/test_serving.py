@@ -408,3 +408,16 @@ def test_serve_cli_fixed_delay_restart(monkeypatch, capsys):
                            "--restartDelay", "10"]) == 0
     assert calls["n"] == 3
     assert "restart 1/3" in capsys.readouterr().out
+
+
+def test_serve_cli_device_flag(tmp_path):
+    """--device pins the store device (cpu here); --spreadShards maps shard
+    s to cuda:(s mod gpus) on GPU nodes (no-op without GPUs)."""
+    from flink_ms_amd.cli.serve import build_app
+    from flink_ms_amd.utils.params import Params
+
+    app = build_app(Params({"device": "cpu"}))
+    assert app.state.als.device.type == "cpu"
+    c = TestClient(app)
+    c.post("/model/als/rows", json={"rows": ["1,U,0.5;0.5"]})
+    assert c.get("/state/ALS_MODEL/1-U").status_code == 200
