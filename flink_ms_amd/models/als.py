@@ -165,17 +165,23 @@ class ALSTrainer:
         t0 = time.perf_counter()
         # C1: item factors to every rank (routed a2a-v or all-gather),
         # then solve local users
+        # the kernels write the next half-iteration's bf16 shard image
+        # directly; any other shard dtype (e.g. fp32 CPU parity configs on
+        # a GPU box) goes through the explicit fp32 copy instead
+        bf16_out = (ctx.device.type == "cuda"
+                    and self.cfg.dtype == torch.bfloat16)
+        on_gpu = ctx.device.type == "cuda"
         item_full = (self.item_route.exchange(self.item_shard)
                      if self.item_route is not None
                      else allgather_rows(ctx, self.item_shard, self.num_items))
         self.user_f32 = ops.als_solve_side(
             self.user_csr, item_full, self.cfg.lambda_,
             out_bf16=self.user_shard[: self.user_csr.num_rows]
-            if ctx.device.type == "cuda" else None,
-            row_order=self.user_order if ctx.device.type == "cuda" else None)
-        if ctx.device.type != "cuda":
+            if bf16_out else None,
+            row_order=self.user_order if on_gpu else None)
+        if not bf16_out:
             self.user_shard[: self.user_csr.num_rows, : self.cfg.num_factors] = (
-                self.user_f32.to(self.cfg.dtype))
+                self.user_f32[:, : self.cfg.num_factors].to(self.cfg.dtype))
         # C1': user factors to every rank, then solve local items
         user_full = (self.user_route.exchange(self.user_shard)
                      if self.user_route is not None
@@ -183,11 +189,11 @@ class ALSTrainer:
         self.item_f32 = ops.als_solve_side(
             self.item_csr, user_full, self.cfg.lambda_,
             out_bf16=self.item_shard[: self.item_csr.num_rows]
-            if ctx.device.type == "cuda" else None,
-            row_order=self.item_order if ctx.device.type == "cuda" else None)
-        if ctx.device.type != "cuda":
+            if bf16_out else None,
+            row_order=self.item_order if on_gpu else None)
+        if not bf16_out:
             self.item_shard[: self.item_csr.num_rows, : self.cfg.num_factors] = (
-                self.item_f32.to(self.cfg.dtype))
+                self.item_f32[:, : self.cfg.num_factors].to(self.cfg.dtype))
         if ctx.device.type == "cuda":
             torch.cuda.synchronize()
         dt = ctx.max_scalar(time.perf_counter() - t0)

@@ -74,9 +74,13 @@ void check_t(const torch::Tensor& t, torch::ScalarType dt, const char* name) {
 }
 
 const unsigned short* bf16_ptr(const torch::Tensor& t) {
+    TORCH_CHECK(t.scalar_type() == torch::kBFloat16,
+                "expected a bf16 tensor at the kernel boundary");
     return reinterpret_cast<const unsigned short*>(t.data_ptr());
 }
 unsigned short* bf16_ptr_mut(torch::Tensor& t) {
+    TORCH_CHECK(t.scalar_type() == torch::kBFloat16,
+                "expected a bf16 output tensor at the kernel boundary");
     return reinterpret_cast<unsigned short*>(t.data_ptr());
 }
 

@@ -80,7 +80,12 @@ def init_from_env(backend: Optional[str] = None,
     rank = int(os.environ["RANK"])
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        # RCCL needs one GPU per local rank; oversubscribed launches (e.g.
+        # a 2-rank CPU-parity run on a 1-GPU box) fall back to gloo
+        local_world = int(os.environ.get(
+            "LOCAL_WORLD_SIZE", os.environ.get("WORLD_SIZE", "1")))
+        backend = ("nccl" if torch.cuda.is_available()
+                   and torch.cuda.device_count() >= local_world else "gloo")
     if backend == "nccl":
         torch.cuda.set_device(local_rank)
         device = torch.device(f"cuda:{local_rank}")
