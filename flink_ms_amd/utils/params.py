@@ -21,6 +21,15 @@ _TRUE = {"true", "1", "yes", "y"}
 _FALSE = {"false", "0", "no", "n"}
 
 
+def _is_number(tok: str) -> bool:
+    """Negative numbers (e.g. ``--bias -0.5``) are values, not flags."""
+    try:
+        float(tok)
+        return True
+    except ValueError:
+        return False
+
+
 class Params:
     """``--key value`` flag map with typed getters (ParameterTool semantics)."""
 
@@ -38,8 +47,10 @@ class Params:
             key = tok.lstrip("-")
             if not key:
                 raise ParamsError("empty flag name")
-            if i + 1 < len(args) and not args[i + 1].startswith("-"):
-                data[key] = args[i + 1]
+            nxt = args[i + 1] if i + 1 < len(args) else None
+            if nxt is not None and (not nxt.startswith("-")
+                                    or _is_number(nxt)):
+                data[key] = nxt
                 i += 2
             else:
                 # bare flag == boolean true (ParameterTool "no value" behavior)

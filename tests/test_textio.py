@@ -109,3 +109,14 @@ def test_svm_range_row_roundtrip_property(pairs):
     b, parsed = t.parse_svm_range_row(row)
     assert b == 7
     assert parsed == [(i, float(w)) for i, w in pairs]
+
+
+def test_params_negative_number_values():
+    from flink_ms_amd.utils.params import Params
+
+    p = Params.from_args(["--lambda", "-0.5", "--seed", "-7", "--flag",
+                          "--name", "x"])
+    assert p.get_float("lambda") == -0.5
+    assert p.get_int("seed") == -7
+    assert p.get_bool("flag") is True
+    assert p.get("name") == "x"
