@@ -111,3 +111,56 @@ def test_bench_json_contract(capsys):
     assert out["higher_is_better"] is True and out["scaling"] == "weak"
     assert out["data"] == "synthetic" and out["value"] > 0
     assert out["config"]["global_batch"] == 1500 * out["n_gpus"]
+
+
+def test_ratings_loader_delimiters(tmp_path):
+    """ALSImpl flag vocabulary: --fieldDelimiter comma|tab,
+    --ignoreFirstLine (ALSImpl.scala:22-32)."""
+    from flink_ms_amd.data.ratings import load_ratings_csv
+
+    tab = tmp_path / "tab.tsv"
+    tab.write_text("1\t2\t3.5\n4\t5\t2.0\n")
+    u, i, r = load_ratings_csv(str(tab), field_delimiter="tab",
+                               ignore_first_line=False)
+    assert u.tolist() == [1, 4] and i.tolist() == [2, 5]
+    assert r.tolist() == [3.5, 2.0]
+
+    com = tmp_path / "c.csv"
+    com.write_text("uId,iId,r\n7,8,1.0\n")
+    u, i, r = load_ratings_csv(str(com))  # defaults: comma + skip header
+    assert u.tolist() == [7] and r.tolist() == [1.0]
+
+
+def test_sharded_routing_unit():
+    """shard_of / row-key routing is deterministic and ingest partitions
+    rows by their state key's shard (no live servers needed)."""
+    from flink_ms_amd.serving.sharding import (ShardedQueryClient,
+                                               als_row_key, shard_of,
+                                               svm_row_key)
+
+    assert als_row_key("42,U,0.5;0.5") == "42-U"
+    assert svm_row_key("17,0.25") == "17"
+    assert shard_of("42-U", 1) == 0
+    s4 = shard_of("42-U", 4)
+    assert 0 <= s4 < 4 and shard_of("42-U", 4) == s4  # stable
+
+    class FakeClient:
+        def __init__(self):
+            self.rows = []
+
+        def ingest_rows(self, model, rows):
+            self.rows.extend(rows)
+            return len(rows)
+
+        def close(self):
+            pass
+
+    sc = ShardedQueryClient.__new__(ShardedQueryClient)
+    sc.clients = [FakeClient() for _ in range(4)]
+    sc.n = 4
+    rows = [f"{i},U,0.1;0.2" for i in range(50)]
+    assert sc.ingest_rows("als", rows) == 50
+    for sh, c in enumerate(sc.clients):
+        for row in c.rows:
+            assert shard_of(als_row_key(row), 4) == sh
+    assert sum(len(c.rows) for c in sc.clients) == 50
