@@ -162,3 +162,34 @@ def test_distributed_als_world4_routed():
         assert p.exitcode == 0
     ids = [i for r in sorted(results) for i in results[r]["user_ids"]]
     assert sorted(ids) == list(range(120))  # every user solved exactly once
+
+
+def _exchange_edge_worker(rank, world, port, q):
+    """One rank contributes ZERO ratings and one owner receives none for
+    some senders — the zero-size all_to_all splits the 8-GPU run can hit."""
+    ctx = _init(rank, world, port)
+    from flink_ms_amd.parallel.shard import Partition, exchange_ratings_by_owner
+
+    part = Partition(total=10, world=world)
+    if rank == 0:
+        keys = torch.tensor([], dtype=torch.int64)
+        other = torch.tensor([], dtype=torch.int64)
+        vals = torch.tensor([], dtype=torch.float32)
+    else:
+        # every rating here is owned by rank 0's range [0,5)
+        keys = torch.tensor([0, 1, 2, 3], dtype=torch.int64)
+        other = torch.tensor([9, 8, 7, 6], dtype=torch.int64)
+        vals = torch.tensor([1.0, 2.0, 3.0, 4.0])
+    k, o, v = exchange_ratings_by_owner(ctx, part, keys, other, vals)
+    q.put((rank, {"k": k.tolist(), "o": o.tolist(), "v": v.tolist()}))
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_exchange_zero_size_splits():
+    res = _run_workers(_exchange_edge_worker)
+    # rank 0 owns keys 0..4: receives all 4 triples (sent by rank 1);
+    # rank 1 owns keys 5..9: receives nothing
+    assert sorted(res[0]["k"]) == [0, 1, 2, 3]
+    assert sorted(res[0]["v"]) == [1.0, 2.0, 3.0, 4.0]
+    assert res[1]["k"] == [] and res[1]["v"] == []
