@@ -193,3 +193,32 @@ def test_exchange_zero_size_splits():
     assert sorted(res[0]["k"]) == [0, 1, 2, 3]
     assert sorted(res[0]["v"]) == [1.0, 2.0, 3.0, 4.0]
     assert res[1]["k"] == [] and res[1]["v"] == []
+
+
+def _routed_empty_worker(rank, world, port, q):
+    """RoutedExchange where one rank requests NOTHING from the other
+    (empty per-pair request lists -> zero-size routed all-to-all-v)."""
+    ctx = _init(rank, world, port)
+    from flink_ms_amd.parallel.routing import RoutedExchange
+    from flink_ms_amd.parallel.shard import Partition
+
+    part = Partition(total=8, world=world)   # rank0 owns 0..3, rank1 4..7
+    # rank 0 references only its own rows; rank 1 references only rank 0's
+    needed = (torch.tensor([1, 2], dtype=torch.int64) if rank == 0
+              else torch.tensor([0, 3], dtype=torch.int64))
+    route = RoutedExchange(ctx, part, needed)
+    shard = (torch.arange(4, dtype=torch.float32).repeat_interleave(4)
+             .reshape(4, 4) + 10 * rank).to(torch.bfloat16)
+    got = route.exchange(shard)
+    remap = route.remap_indices(needed)
+    vals = got[remap][:, 0].to(torch.float32)
+    q.put((rank, vals.tolist()))
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_routed_exchange_empty_requests():
+    res = _run_workers(_routed_empty_worker)
+    # row r of rank R's shard has leading value 10R + r
+    assert res[0] == [1.0, 2.0]   # own rows 1,2
+    assert res[1] == [0.0, 3.0]   # rank 0's rows 0,3
