@@ -99,6 +99,14 @@ def plan_exchange(ctx: DistContext, part: Partition,
            else torch.device("cpu"))
     needed = torch.unique(csr_indices.to(dev).long())
     frac = needed.numel() / max(1, part.total)
+    if mode == "auto":
+        # the routed/allgather choice must be GLOBALLY uniform: a rank
+        # entering RoutedExchange's collectives while another skips them
+        # deadlocks the job.  MAX over ranks: if any rank is dense, all
+        # fall back to the single bucketed all-gather.
+        frac_t = torch.tensor([frac], dtype=torch.float64, device=dev)
+        dist.all_reduce(frac_t, op=dist.ReduceOp.MAX)
+        frac = float(frac_t.item())
     if mode == "auto" and frac >= dense_threshold:
         return None, None
     route = RoutedExchange(ctx, part, needed)

@@ -222,3 +222,27 @@ def test_routed_exchange_empty_requests():
     # row r of rank R's shard has leading value 10R + r
     assert res[0] == [1.0, 2.0]   # own rows 1,2
     assert res[1] == [0.0, 3.0]   # rank 0's rows 0,3
+
+
+def _plan_disagree_worker(rank, world, port, q):
+    """Ranks whose LOCAL referenced fractions straddle the dense threshold
+    must still make a UNIFORM routed/allgather choice (MAX-reduced), or the
+    job deadlocks with one rank inside RoutedExchange's collectives."""
+    ctx = _init(rank, world, port)
+    from flink_ms_amd.parallel.routing import plan_exchange
+    from flink_ms_amd.parallel.shard import Partition
+
+    part = Partition(total=100, world=world)
+    # rank 0: references 10% of the side; rank 1: 90% (>= 0.7 threshold)
+    idx = (torch.arange(10, dtype=torch.int32) if rank == 0
+           else torch.arange(90, dtype=torch.int32))
+    route, remap = plan_exchange(ctx, part, idx, mode="auto")
+    q.put((rank, route is None))
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_plan_exchange_uniform_decision():
+    res = _run_workers(_plan_disagree_worker)
+    # MAX(0.1, 0.9) = 0.9 >= 0.7 -> BOTH ranks take the allgather path
+    assert res[0] is True and res[1] is True
