@@ -171,21 +171,38 @@ def create_app(als_store: Optional[ALSModelStore] = None,
 
     # ----------------------------------------------------------- ingest
 
+    # Malformed model rows are a CLIENT error (400 + the parse message);
+    # the reference's analog is a consumer map() throw that fails/restarts
+    # the whole serving job -- an HTTP surface reports instead of dying.
+    def _ingest_or_400(store, rows):
+        try:
+            return {"ingested": store.ingest(rows)}
+        except (ValueError, IndexError) as e:
+            raise HTTPException(400, f"malformed model row: {e}")
+
     @app.post("/model/als/rows")
     def als_rows(body: RowsBody):
-        return {"ingested": als.ingest(body.rows)}
+        return _ingest_or_400(als, body.rows)
 
     @app.post("/model/als/load")
     def als_load(body: LoadBody):
-        return {"ingested": als.ingest(_read_rows(body.path))}
+        try:
+            rows = _read_rows(body.path)
+        except OSError as e:
+            raise HTTPException(400, f"cannot read model path: {e}")
+        return _ingest_or_400(als, rows)
 
     @app.post("/model/svm/rows")
     def svm_rows(body: RowsBody):
-        return {"ingested": svm.ingest(body.rows)}
+        return _ingest_or_400(svm, body.rows)
 
     @app.post("/model/svm/load")
     def svm_load(body: LoadBody):
-        return {"ingested": svm.ingest(_read_rows(body.path))}
+        try:
+            rows = _read_rows(body.path)
+        except OSError as e:
+            raise HTTPException(400, f"cannot read model path: {e}")
+        return _ingest_or_400(svm, rows)
 
     @app.post("/checkpoint")
     def checkpoint():

@@ -421,3 +421,14 @@ def test_serve_cli_device_flag(tmp_path):
     c = TestClient(app)
     c.post("/model/als/rows", json={"rows": ["1,U,0.5;0.5"]})
     assert c.get("/state/ALS_MODEL/1-U").status_code == 200
+
+
+def test_malformed_ingest_is_400(als_store, svm_store):
+    c = TestClient(create_app(als_store, svm_store))
+    r = c.post("/model/als/rows", json={"rows": ["garbage-no-commas"]})
+    assert r.status_code == 400 and "malformed" in r.json()["detail"]
+    r = c.post("/model/als/load", json={"path": "/nonexistent/m.model"})
+    assert r.status_code == 400
+    # a valid batch after the failure still ingests (store not poisoned)
+    assert c.post("/model/als/rows",
+                  json={"rows": ["123,U,1.0;2.0;3.0"]}).json()["ingested"] == 1
