@@ -107,3 +107,43 @@ def test_sdca_reference_converges():
     o1 = R.hinge_objective(csr, y, v, 0.01)
     assert o1 < o0
     assert (alpha >= 0).all() and (alpha <= 1).all()
+
+
+def test_csr_transpose_involution():
+    from flink_ms_amd.data.blocked import csr_from_coo, csr_transpose
+
+    torch.manual_seed(0)
+    rows = torch.randint(0, 50, (400,)).int()
+    cols = torch.randint(0, 30, (400,)).int()
+    vals = torch.randn(400)
+    a = csr_from_coo(rows, cols, vals, 50, 30)
+    tt = csr_transpose(csr_transpose(a))
+    assert tt.num_rows == a.num_rows and tt.num_cols == a.num_cols
+    assert torch.equal(tt.indptr, a.indptr)
+    # same per-row column/value MULTISETS (transpose sorts within rows)
+    for r in range(a.num_rows):
+        s, e = int(a.indptr[r]), int(a.indptr[r + 1])
+        got = sorted(zip(tt.indices[s:e].tolist(), tt.values[s:e].tolist()))
+        want = sorted(zip(a.indices[s:e].tolist(), a.values[s:e].tolist()))
+        assert got == want
+
+
+def test_dense_from_csr_roundtrip():
+    """CSR built from COO reproduces the dense matrix exactly (duplicate
+    (row,col) pairs accumulate in dense but stay distinct entries in CSR,
+    matching what the Gramian contraction sums over)."""
+    from flink_ms_amd.data.blocked import csr_from_coo
+
+    torch.manual_seed(1)
+    rows = torch.randint(0, 8, (60,)).int()
+    cols = torch.randint(0, 6, (60,)).int()
+    vals = torch.randn(60)
+    dense = torch.zeros(8, 6)
+    dense.index_put_((rows.long(), cols.long()), vals, accumulate=True)
+    csr = csr_from_coo(rows, cols, vals, 8, 6)
+    re = torch.zeros(8, 6)
+    for r in range(8):
+        s, e = int(csr.indptr[r]), int(csr.indptr[r + 1])
+        for c, v in zip(csr.indices[s:e].tolist(), csr.values[s:e].tolist()):
+            re[r, c] += v
+    assert torch.allclose(re, dense, atol=1e-6)
