@@ -66,13 +66,18 @@ def main(argv=None) -> int:
         raise RuntimeError("Unable to load User mean or item mean factors.")
     seen = 0
     total = 0
+    polls = 0
+    # --maxPolls: operational bound for tests/drains (not a reference flag;
+    # the reference job streams forever)
+    max_polls = params.get_int("maxPolls", 0)
     while True:
         with open(path) as f:
             rows = [line for line in f.read().splitlines() if line.strip()]
         new_rows = rows[seen:]
         seen = len(rows)
         total += run_once(client, new_rows, params)
-        if mode != "continuous":
+        polls += 1
+        if mode != "continuous" or (max_polls and polls >= max_polls):
             break
         time.sleep(interval_ms / 1000.0)
     print(f"applied {total} online updates")

@@ -262,3 +262,37 @@ def test_driver_bench_launch_contract(tmp_path):
     assert d["config"]["global_batch"] == 10_000  # whole-job aggregate
     assert d["value"] > 0 and d["ms_per_step"] > 0
     assert d["config"]["parallelism"] == "dp2+factor-allgather"
+
+
+@pytest.mark.timeout(120)
+def test_sgd_continuous_mode_polls_appended_rows(tmp_path, server):
+    """--mode continuous re-polls the input for appended rows every
+    --interval ms (SGD.java:48-64 streaming source parity); bounded here
+    via the operational --maxPolls hook."""
+    port = server
+    with QueryClientHelper("127.0.0.1", port, 5) as c:
+        c.ingest_rows("als", ["700001,U,1.0;1.0", "800001,I,1.0;1.0",
+                              "MEAN,U,0.5;0.5", "MEAN,I,0.5;0.5"])
+        before = c.query_state("ALS_MODEL", "700001-U")[1]
+
+    stream = tmp_path / "stream.tsv"
+    stream.write_text("700001\t800001\t5.0\n")
+
+    import threading as th
+    from flink_ms_amd.cli import sgd as sgd_cli
+
+    def appender():
+        time.sleep(0.6)
+        with open(stream, "a") as f:
+            f.write("700001\t800001\t1.0\n")
+    t = th.Thread(target=appender)
+    t.start()
+    rc = sgd_cli.main(["--input", str(stream), "--mode", "continuous",
+                       "--interval", "500", "--maxPolls", "3",
+                       "--jobManagerHost", "127.0.0.1",
+                       "--jobManagerPort", str(port)])
+    t.join()
+    assert rc == 0
+    with QueryClientHelper("127.0.0.1", port, 5) as c:
+        after = c.query_state("ALS_MODEL", "700001-U")[1]
+    assert after != before  # both polls' updates landed
