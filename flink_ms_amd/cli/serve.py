@@ -70,7 +70,7 @@ def main(argv=None) -> int:
                 if n > 0:
                     pd["device"] = f"cuda:{s % n}"
             procs.append(ctx.Process(target=_run_shard,
-                                     args=(pd, base + s)))
+                                     args=(pd, base + s), daemon=True))
         for p in procs:
             p.start()
         print(f"serving {shards} key-partitioned shards on ports "

@@ -296,3 +296,40 @@ def test_sgd_continuous_mode_polls_appended_rows(tmp_path, server):
     with QueryClientHelper("127.0.0.1", port, 5) as c:
         after = c.query_state("ALS_MODEL", "700001-U")[1]
     assert after != before  # both polls' updates landed
+
+
+@pytest.mark.timeout(180)
+def test_serve_cli_shards_subprocess(tmp_path):
+    """The real `serve --shards 2` CLI: two shard processes come up on
+    consecutive ports and serve routed state; shard children are daemonic
+    (die with the job, Flink TaskManager lifetime parity)."""
+    import subprocess
+    import sys
+
+    base = _free_port()
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "flink_ms_amd.cli.serve", "--shards", "2",
+         "--port", str(base), "--device", "cpu"],
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT)
+    try:
+        from flink_ms_amd.serving.sharding import ShardedQueryClient
+        eps = [("127.0.0.1", base), ("127.0.0.1", base + 1)]
+        sc = ShardedQueryClient(eps, 2.0)
+        for c in sc.clients:
+            for _ in range(200):
+                try:
+                    c._client.get(c.base + "/healthz").raise_for_status()
+                    break
+                except Exception:
+                    time.sleep(0.1)
+            else:
+                raise AssertionError("shard did not come up")
+        rows = [f"{i},U,0.5;0.5" for i in range(1, 21)]
+        assert sc.ingest_rows("als", rows) == 20
+        for i in (1, 7, 20):
+            hit = sc.query_state("ALS_MODEL", f"{i}-U")
+            assert hit is not None and hit[1] == "0.5;0.5"
+        sc.close()
+    finally:
+        proc.terminate()
+        proc.wait(timeout=30)
