@@ -147,3 +147,22 @@ def test_dense_from_csr_roundtrip():
         for c, v in zip(csr.indices[s:e].tolist(), csr.values[s:e].tolist()):
             re[r, c] += v
     assert torch.allclose(re, dense, atol=1e-6)
+
+
+def test_libsvm_roundtrip(tmp_path):
+    """write_libsvm -> read_libsvm preserves rows, labels and the 1-BASED
+    index convention (flink-ml readLibSVM parity, SVMImpl.scala:21)."""
+    from flink_ms_amd.data.libsvm import (LibSVMShape, read_libsvm,
+                                          synthetic_libsvm, write_libsvm)
+
+    csr, y = synthetic_libsvm(LibSVMShape(40, 25, 5), seed=9)
+    p = tmp_path / "t.libsvm"
+    write_libsvm(str(p), csr, y)
+    first = open(p).readline().split()
+    assert first[0] in ("+1", "-1", "1", "-1.0", "1.0")
+    assert all(int(tok.split(":")[0]) >= 1 for tok in first[1:])  # 1-based
+    csr2, y2, nfeat = read_libsvm(str(p))
+    assert csr2.num_rows == 40 and torch.equal(y2, y)
+    assert torch.equal(csr2.indptr, csr.indptr)
+    assert torch.equal(csr2.indices, csr.indices)   # back to 0-based
+    assert torch.allclose(csr2.values, csr.values, atol=1e-6)
