@@ -2,6 +2,7 @@
 full data-flow topology: train -> mean vector -> publish ->
 serve -> predict / online SGD / MSE / load generators."""
 
+import os
 import socket
 import threading
 import time
@@ -219,7 +220,8 @@ def test_distributed_cli_training(tmp_path):
     res = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29778", "-m", "flink_ms_amd.cli.als_train",
+         "--master-port", str(29600 + os.getpid() % 300), "-m",
+         "flink_ms_amd.cli.als_train",
          "--input", str(csv), "--iterations", "2", "--numFactors", "8",
          "--lambda", "0.1", "--userFactors", str(uf),
          "--itemFactors", str(if_)],
@@ -248,7 +250,8 @@ def test_driver_bench_launch_contract(tmp_path):
     res = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29779", "bench.py", "--gpus", "2",
+         "--master-port", str(29100 + os.getpid() % 300),
+         "bench.py", "--gpus", "2",
          "--steps", "2", "--warmup", "1", "--device", "cpu",
          "--users-per-gpu", "300", "--items", "200",
          "--ratings-per-gpu", "5000", "--rank", "16"],
