@@ -336,3 +336,35 @@ def test_serve_cli_shards_subprocess(tmp_path):
     finally:
         proc.terminate()
         proc.wait(timeout=30)
+
+
+def test_temporary_path_stages_iterations(tmp_path):
+    """--temporaryPath stages every iteration's factors in the model text
+    format (reference flag, ALSImpl.scala:42-44; here: restartable
+    training instead of a memory trade)."""
+    from flink_ms_amd.cli import als_train
+    from flink_ms_amd.utils.textio import parse_als_row
+
+    shape = RatingsShape(50, 30, 900)
+    u, i, r = synthetic_ratings(shape, seed=11)
+    csv = tmp_path / "r.csv"
+    with open(csv, "w") as f:
+        f.write("h,h,h\n")
+        for a, b, c in zip(u.tolist(), i.tolist(), r.tolist()):
+            f.write(f"{a},{b},{c}\n")
+    stage = tmp_path / "stage"
+    rc = als_train.main(["--input", str(csv), "--iterations", "3",
+                         "--numFactors", "8",
+                         "--temporaryPath", str(stage),
+                         "--userFactors", str(tmp_path / "uf"),
+                         "--itemFactors", str(tmp_path / "if")])
+    assert rc == 0
+    dirs = sorted(p.name for p in stage.iterdir())
+    assert dirs == ["iteration-0", "iteration-1", "iteration-2"]
+    rows = open(stage / "iteration-1" / "userFactors").read().splitlines()
+    assert len(rows) == 50
+    _, kind, fac = parse_als_row(rows[0])
+    assert kind == "U" and len(fac) == 8
+    # the staged final iteration matches the written model
+    assert open(stage / "iteration-2" / "userFactors").read() == \
+        open(tmp_path / "uf").read()
