@@ -432,3 +432,26 @@ def test_malformed_ingest_is_400(als_store, svm_store):
     # a valid batch after the failure still ingests (store not poisoned)
     assert c.post("/model/als/rows",
                   json={"rows": ["123,U,1.0;2.0;3.0"]}).json()["ingested"] == 1
+
+
+def test_periodic_checkpoint_thread(tmp_path, als_store):
+    """The interval checkpointer writes snapshots on its own (consumer
+    parity: enableCheckpointing(interval), ALSKafkaConsumer.java:44-47)."""
+    import glob as glob_mod
+
+    uri = str(tmp_path / "ckpt")
+    app = create_app(als_store, checkpoint_data_uri=uri,
+                     checkpoint_interval_ms=60)
+    try:
+        deadline = time.time() + 10
+        while time.time() < deadline:
+            snaps = glob_mod.glob(uri + "/als-*.model")
+            if snaps:
+                break
+            time.sleep(0.05)
+        assert snaps, "no periodic snapshot appeared"
+        restored = ALSModelStore(device=torch.device("cpu"))
+        restored.ingest(open(snaps[0]).read().splitlines())
+        assert restored.query("1-U") == als_store.query("1-U")
+    finally:
+        app.state._ckpt_stop.set()
