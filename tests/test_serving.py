@@ -455,3 +455,15 @@ def test_periodic_checkpoint_thread(tmp_path, als_store):
         assert restored.query("1-U") == als_store.query("1-U")
     finally:
         app.state._ckpt_stop.set()
+
+
+def test_rest_error_codes(als_store, svm_store):
+    c = TestClient(create_app(als_store, svm_store))
+    r = c.post("/als/predict_batch", json={"users": ["1", "2"],
+                                           "items": ["10"]})
+    assert r.status_code == 400  # length mismatch
+    assert c.get("/state/NOPE_MODEL/1").status_code == 404
+    # SGD on an unknown pair without MEAN rows present -> 400, not a crash
+    empty = TestClient(create_app())
+    r = empty.post("/sgd/update", json={"ratings": ["5\t6\t3.0"]})
+    assert r.status_code == 400
