@@ -161,8 +161,13 @@ class SVMTrainer:
         dw = v - self.w
         ctx.all_reduce_(dw)
         self.w += (cfg.stepsize / K) * dw
-        # safe CoCoA averaging of the duals
-        self.alpha = a0 + (self.alpha - a0) / K
+        # safe CoCoA averaging of the duals — IN PLACE: the captured hipGraph
+        # (and the kernels inside it) keep reading/writing THIS alpha buffer,
+        # so a rebind would silently divorce the averaged duals from the
+        # buffer the SDCA kernel updates (K>1 would then run CoCoA-add
+        # semantics on w but full local duals — wrong primal/dual pairing)
+        if K > 1:
+            self.alpha.sub_(a0).div_(K).add_(a0)
         if ctx.device.type == "cuda":
             torch.cuda.synchronize()
         dt = ctx.max_scalar(time.perf_counter() - t0)

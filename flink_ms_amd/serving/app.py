@@ -252,10 +252,14 @@ def create_app(als_store: Optional[ALSModelStore] = None,
             row = row.strip()
             if not row:
                 continue
-            u, i, r = row.split(body.field_delimiter)[:3]
+            try:
+                u, i, r = row.split(body.field_delimiter)[:3]
+                rating = float(r)
+            except ValueError:
+                raise HTTPException(400, f"malformed rating row: {row!r}")
             try:
                 rows = als.sgd_update(
-                    u, i, float(r), body.learning_rate,
+                    u, i, rating, body.learning_rate,
                     body.user_regularization, body.item_regularization,
                     body.user_mean, body.item_mean,
                     v0_semantics=body.v0_semantics)
@@ -279,12 +283,16 @@ def create_app(als_store: Optional[ALSModelStore] = None,
             row = row.strip()
             if not row:
                 continue
-            u, i, r = row.split(body.field_delimiter)[:3]
+            try:
+                u, i, r = row.split(body.field_delimiter)[:3]
+                rating = float(r)
+            except ValueError:
+                raise HTTPException(400, f"malformed rating row: {row!r}")
             pred = als.predict(u, i)
             if pred is None:
                 skipped += 1
                 continue
-            se_sum += (float(r) - pred) ** 2
+            se_sum += (rating - pred) ** 2
             count += 1
         return {"mse": (se_sum / count) if count else None,
                 "scored": count, "skipped": skipped}

@@ -293,12 +293,34 @@ class ALSModelStore:
             return list(self._payload.keys())
 
     def snapshot_rows(self) -> List[str]:
-        """All state as model-format text rows (the checkpoint format)."""
+        """All state as model-format text rows (the checkpoint format).
+
+        Covers BOTH explicitly ingested payloads and tensor-attached factors
+        (attach_factors): the reference's checkpoint spans all keyed state
+        (ALSKafkaConsumer.java:44-46 enableCheckpointing on the whole
+        consumer job), so a snapshot taken after an in-process train->serve
+        handoff must persist the full model, not just the lazily queried
+        keys.  Ingested rows win over attached rows (hot-swap contract)."""
         with self._lock:
             out = []
             for key, payload in self._payload.items():
                 entity_id, kind = key.rsplit("-", 1)
                 out.append(f"{entity_id},{kind},{payload}")
+            att = self._attached
+            if att is not None:
+                for kind in ("U", "I"):
+                    fac, idmap = att.get(kind, (None, None))
+                    if fac is None:
+                        continue
+                    ids = (sorted(idmap.items(), key=lambda kv: kv[1])
+                           if idmap is not None
+                           else [(r, r) for r in range(fac.shape[0])])
+                    for entity_id, row in ids:
+                        key = als_state_key(entity_id, kind)
+                        if key in self._payload:
+                            continue
+                        out.append(f"{entity_id},{kind},"
+                                   + self._fmt(fac[row].tolist()))
             return out
 
     def __len__(self) -> int:

@@ -21,6 +21,15 @@ semantics: shortest decimal that round-trips, decimal notation for
 1e-3 <= |d| < 1e7, computerized scientific notation otherwise).  Python's
 ``repr`` provides the same shortest-round-trip digit selection; only the
 surface formatting differs, which is what this module implements.
+
+PARITY SCOPE: byte parity is guaranteed against JDK>=19 ``Double.toString``
+(Ryu, JDK-4511638).  The reference stack is Flink-1.3-era Java 8, whose
+legacy ``FloatingDecimal`` emits LONGER (non-shortest) digit strings for
+some doubles — rows produced by an actual Java 8 reference job can differ
+in those rare trailing digits.  Values still round-trip identically (every
+consumer, reference included, re-parses with ``Double.parseDouble``), so
+the numeric state contract holds; only byte-for-byte diffing of Java-8-
+produced files against ours can show benign differences.
 """
 
 from __future__ import annotations

@@ -467,3 +467,30 @@ def test_rest_error_codes(als_store, svm_store):
     empty = TestClient(create_app())
     r = empty.post("/sgd/update", json={"ratings": ["5\t6\t3.0"]})
     assert r.status_code == 400
+
+
+def test_snapshot_covers_attached_factors():
+    """Checkpoints after an in-process attach must persist the WHOLE model,
+    not just the lazily queried keys (reference: the consumer checkpoints
+    all keyed state, ALSKafkaConsumer.java:44-46)."""
+    from flink_ms_amd.data.ratings import RatingsShape, synthetic_ratings
+    from flink_ms_amd.models.als import ALSConfig, train_als
+    u, i, r = synthetic_ratings(RatingsShape(30, 15, 400), seed=8)
+    model, _ = train_als(u, i, r, 30, 15,
+                         ALSConfig(iterations=2, num_factors=8,
+                                   lambda_=0.1, dtype=torch.float32))
+    store = ALSModelStore(device=torch.device("cpu"))
+    store.attach_factors(model.user_factors, model.item_factors,
+                         model.user_ids, model.item_ids)
+    store.query("3-U")                       # touch one key only
+    store.ingest_row("3,U,9.0;0;0;0;0;0;0;0")  # hot-swap over attached
+    rows = store.snapshot_rows()
+    assert len(rows) == 30 + 15              # full model, no duplicates
+    restored = ALSModelStore(device=torch.device("cpu"))
+    restored.ingest(rows)
+    # hot-swapped row won over the attached value
+    assert restored.query("3-U")[1].startswith("9.0")
+    # an untouched key restores to the same payload the live store serves
+    assert restored.query("7-I") == store.query("7-I")
+    assert restored.predict("5", "2") == pytest.approx(
+        store.predict("5", "2"), rel=1e-12)
