@@ -254,7 +254,8 @@ def test_driver_bench_launch_contract(tmp_path):
          "bench.py", "--gpus", "2",
          "--steps", "2", "--warmup", "1", "--device", "cpu",
          "--users-per-gpu", "300", "--items", "200",
-         "--ratings-per-gpu", "5000", "--rank", "16"],
+         "--ratings-per-gpu", "5000", "--rank", "16",
+         "--svm-rows-per-gpu", "400"],
         capture_output=True, text=True, timeout=240)
     assert res.returncode == 0, res.stderr[-2000:]
     json_lines = [ln for ln in res.stdout.splitlines()

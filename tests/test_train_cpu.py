@@ -88,7 +88,8 @@ def test_bench_cpu_smoke():
     import bench
     rc = bench.main(["--device", "cpu", "--steps", "1", "--warmup", "0",
                      "--rank", "8", "--ratings-per-gpu", "2000",
-                     "--users-per-gpu", "150", "--items", "80"])
+                     "--users-per-gpu", "150", "--items", "80",
+                     "--svm-rows-per-gpu", "300"])
     assert rc == 0
 
 
@@ -99,7 +100,8 @@ def test_bench_json_contract(capsys):
     import bench
     rc = bench.main(["--device", "cpu", "--steps", "1", "--warmup", "0",
                      "--rank", "8", "--ratings-per-gpu", "1500",
-                     "--users-per-gpu", "100", "--items", "60"])
+                     "--users-per-gpu", "100", "--items", "60",
+                     "--svm-rows-per-gpu", "250"])
     assert rc == 0
     line = [ln for ln in capsys.readouterr().out.splitlines()
             if ln.startswith("{")][-1]
@@ -111,6 +113,9 @@ def test_bench_json_contract(capsys):
     assert out["higher_is_better"] is True and out["scaling"] == "weak"
     assert out["data"] == "synthetic" and out["value"] > 0
     assert out["config"]["global_batch"] == 1500 * out["n_gpus"]
+    # the CoCoA-SVM secondary metric rides in config BY DEFAULT (no --svm)
+    assert out["config"]["svm_samples_per_sec"] > 0
+    assert out["config"]["timed_region_s"] > 0
 
 
 def test_ratings_loader_delimiters(tmp_path):
