@@ -49,6 +49,12 @@ def parse_args(argv=None):
     p.add_argument("--users-per-gpu", type=int, default=ML25M_SHAPE.num_users)
     p.add_argument("--items", type=int, default=ML25M_SHAPE.num_items)
     p.add_argument("--lambda", dest="lambda_", type=float, default=0.9)
+    p.add_argument("--factor-dtype", choices=["fp8", "bf16"], default="fp8",
+                   help="factor storage/exchange precision on GPU (fp8 = "
+                        "OCP e4m3 factor gathers + exchange, fp32 normal "
+                        "equations/solve; ~1-2%% relative train-MSE cost, "
+                        "benchmarks/fp8_convergence_study.py).  CPU runs "
+                        "always use fp32.")
     p.add_argument("--seed", type=int, default=42)
     p.add_argument("--scale-1b", action="store_true",
                    help="1B-rating config: 125M ratings x 1.25M users per GPU")
@@ -82,10 +88,11 @@ def bench_als(args, ctx):
     # variance averages out; the reported value stays PER ITERATION and
     # every iteration does identical full work (solve both sides)
     ips = args.iters_per_step or (10 if ctx.device.type == "cuda" else 1)
+    on_gpu = ctx.device.type == "cuda"
     cfg = ALSConfig(iterations=args.steps * ips, num_factors=args.rank,
                     lambda_=args.lambda_, seed=args.seed,
-                    dtype=torch.bfloat16 if ctx.device.type == "cuda"
-                    else torch.float32)
+                    dtype=torch.bfloat16 if on_gpu else torch.float32,
+                    factor_dtype=args.factor_dtype if on_gpu else "bf16")
     trainer = ALSTrainer(cfg, ctx)
     trainer.setup(u, i.long(), r, num_users, items)
 
@@ -120,6 +127,10 @@ def bench_als(args, ctx):
             "iterations_per_step": ips,
             "ms_per_iteration": elapsed / (args.steps * ips) * 1000.0,
             "timed_region_s": elapsed,
+            "factor_exchange": ("fp8-e4m3" if on_gpu
+                                and args.factor_dtype == "fp8" else
+                                ("bf16" if on_gpu else "fp32")),
+            "solve_dtype": "fp32",
             "parallelism": f"dp{world}+factor-allgather",
         },
     }
@@ -179,7 +190,9 @@ def main(argv=None):
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,  # reference publishes no numbers
-            "dtype": "bf16" if ctx.device.type == "cuda" else "fp32",
+            "dtype": (("bf16+fp8-factors" if args.factor_dtype == "fp8"
+                       else "bf16")
+                      if ctx.device.type == "cuda" else "fp32"),
             "data": "synthetic",
             "config": {**als["config"], **extra},
         }

@@ -45,6 +45,12 @@ class ALSConfig:
     seed: int = 42
     # compute dtype of the factor operands fed to the Gramian kernels
     dtype: torch.dtype = torch.bfloat16
+    # factor storage/exchange precision: 'bf16', or 'fp8' = OCP e4m3 bytes —
+    # one cache line per gathered k<=64 row in the (gather-latency-bound)
+    # Gramian and half the xGMI exchange bytes, at ~1-2% relative train-MSE
+    # cost (benchmarks/fp8_convergence_study.py).  Normal equations and the
+    # LDL solve stay fp32 either way.
+    factor_dtype: str = 'bf16'
     # C1 exchange: 'auto' picks routed all-to-all-v when the referenced
     # fraction of the opposite side is sparse, full all-gather otherwise
     routed_exchange: str = 'auto'
@@ -139,14 +145,17 @@ class ALSTrainer:
         k = self.cfg.num_factors
         kp = ((k + 15) // 16) * 16 if dev.type == "cuda" else k
         self._kp = kp
+        self._fp8 = self.cfg.factor_dtype == "fp8"
+        shard_dtype = torch.uint8 if self._fp8 else self.cfg.dtype
         # local shards are padded to uniform size for all_gather_into_tensor
         self.item_shard = torch.zeros(self.ipart.shard_size, kp,
-                                      dtype=self.cfg.dtype, device=dev)
+                                      dtype=shard_dtype, device=dev)
         init = _init_factors(ihi - ilo, k, self.cfg.seed + ctx.rank, dev,
-                             self.cfg.dtype)
-        self.item_shard[: ihi - ilo, :k] = init
+                             torch.float32)
+        self.item_shard[: ihi - ilo, :k] = (
+            ops.quantize_fp8(init) if self._fp8 else init.to(self.cfg.dtype))
         self.user_shard = torch.zeros(self.upart.shard_size, kp,
-                                      dtype=self.cfg.dtype, device=dev)
+                                      dtype=shard_dtype, device=dev)
         self.user_f32: Optional[torch.Tensor] = None
         self.item_f32: Optional[torch.Tensor] = None
         log.info(
@@ -168,9 +177,11 @@ class ALSTrainer:
         # the kernels write the next half-iteration's bf16 shard image
         # directly; any other shard dtype (e.g. fp32 CPU parity configs on
         # a GPU box) goes through the explicit fp32 copy instead
-        bf16_out = (ctx.device.type == "cuda"
-                    and self.cfg.dtype == torch.bfloat16)
         on_gpu = ctx.device.type == "cuda"
+        fp8 = self._fp8
+        bf16_out = (on_gpu and not fp8
+                    and self.cfg.dtype == torch.bfloat16)
+        direct_out = bf16_out or fp8   # solver writes the shard image itself
         item_full = (self.item_route.exchange(self.item_shard)
                      if self.item_route is not None
                      else allgather_rows(ctx, self.item_shard, self.num_items))
@@ -178,8 +189,10 @@ class ALSTrainer:
             self.user_csr, item_full, self.cfg.lambda_,
             out_bf16=self.user_shard[: self.user_csr.num_rows]
             if bf16_out else None,
+            out_fp8=self.user_shard[: self.user_csr.num_rows]
+            if fp8 else None,
             row_order=self.user_order if on_gpu else None)
-        if not bf16_out:
+        if not direct_out:
             self.user_shard[: self.user_csr.num_rows, : self.cfg.num_factors] = (
                 self.user_f32[:, : self.cfg.num_factors].to(self.cfg.dtype))
         # C1': user factors to every rank, then solve local items
@@ -190,8 +203,10 @@ class ALSTrainer:
             self.item_csr, user_full, self.cfg.lambda_,
             out_bf16=self.item_shard[: self.item_csr.num_rows]
             if bf16_out else None,
+            out_fp8=self.item_shard[: self.item_csr.num_rows]
+            if fp8 else None,
             row_order=self.item_order if on_gpu else None)
-        if not bf16_out:
+        if not direct_out:
             self.item_shard[: self.item_csr.num_rows, : self.cfg.num_factors] = (
                 self.item_f32[:, : self.cfg.num_factors].to(self.cfg.dtype))
         if ctx.device.type == "cuda":

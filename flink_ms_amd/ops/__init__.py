@@ -63,6 +63,30 @@ def _empty(device) -> torch.Tensor:
     return _EMPTY[key]
 
 
+# ------------------------------------------------------------- fp8 e4m3
+
+def quantize_fp8(t: torch.Tensor) -> torch.Tensor:
+    """fp32/bf16 -> OCP e4m3fn bytes (uint8).  The byte image is what the
+    fp8 Gramian kernels gather (one cache line per k<=64 factor row) and
+    what the factor exchange ships over xGMI (half the bf16 bytes)."""
+    return t.to(torch.float8_e4m3fn).view(torch.uint8)
+
+
+def dequantize_fp8(t: torch.Tensor) -> torch.Tensor:
+    """uint8 e4m3 bytes -> fp32 (exact: every e4m3 value is a float)."""
+    return t.view(torch.float8_e4m3fn).to(torch.float32)
+
+
+def fp8_rating_pair(r: torch.Tensor) -> torch.Tensor:
+    """The EXT-column rating representation of the fp8 Gramian: e4m3 hi/lo
+    pair with r ~= hi + lo (max relative error ~1.9e-3, see
+    benchmarks/fp8_convergence_study.py).  Returns the dequantized
+    effective rating, for CPU references that must match the kernel."""
+    hi = dequantize_fp8(quantize_fp8(r.float()))
+    lo = dequantize_fp8(quantize_fp8(r.float() - hi))
+    return hi + lo
+
+
 # -------------------------------------------------------------------- ALS
 
 def als_solve_side(
@@ -70,32 +94,48 @@ def als_solve_side(
     other_factors: torch.Tensor,
     reg: float,
     out_bf16: Optional[torch.Tensor] = None,
+    out_fp8: Optional[torch.Tensor] = None,
     row_order: Optional[torch.Tensor] = None,
     fused: bool = False,
     slab_rows: Optional[int] = None,
 ) -> torch.Tensor:
     """One ALS half-iteration: solve every row entity of ``csr`` against the
     opposite side's factors.  Returns fp32 [num_rows, k]; optionally also
-    writes the bf16 image for the next half-iteration.
+    writes the bf16 (or e4m3 ``out_fp8``) image for the next half-iteration.
 
-    Two GPU paths: the default modular path (MFMA Gramian kernel -> batched
+    GPU paths: the default modular path (MFMA Gramian kernel -> batched
     wave-per-entity LDL solve; best measured occupancy at k <= 64) and the
-    single-launch fused kernel (``fused=True``; avoids the A round-trip
-    through HBM but serializes the solve at Gramian occupancy).
+    single-launch fused kernel (``fused=True``/k > 64; avoids the A
+    round-trip through HBM but serializes the solve at Gramian occupancy).
+    ``other_factors`` dtype selects the gather precision: uint8 = e4m3
+    bytes (one cache line per k<=64 row — the r2 headline path), anything
+    else is cast to bf16.
     """
     if other_factors.is_cuda:
         ops = _require_hip()
-        fac = _pad_k(other_factors.to(torch.bfloat16).contiguous())
+        fp8 = other_factors.dtype == torch.uint8
+        if fp8:
+            fac = other_factors.contiguous()   # pre-padded e4m3 byte image
+            if fac.shape[1] % 16:
+                raise ValueError("fp8 factor image must be 16-col padded")
+        else:
+            fac = _pad_k(other_factors.to(torch.bfloat16).contiguous())
         k = fac.shape[1]
         ob = out_bf16 if out_bf16 is not None else _empty(fac.device)
+        o8 = out_fp8 if out_fp8 is not None else _empty(fac.device)
         ro = row_order if row_order is not None else _empty(fac.device)
         korig = other_factors.shape[1]
         if fused or k > 64:
             # k>64: fused wins (no nrows*k*k A round-trip through HBM)
             out = torch.empty(csr.num_rows, k, dtype=torch.float32,
                               device=fac.device)
-            ops.als_solve_fused(csr.indptr, csr.indices, csr.values, fac,
-                                out, ob, ro, float(reg), _stream())
+            if fp8:
+                ops.als_solve_fused_fp8(csr.indptr, csr.indices, csr.values,
+                                        fac, out, o8, ro, float(reg),
+                                        _stream())
+            else:
+                ops.als_solve_fused(csr.indptr, csr.indices, csr.values, fac,
+                                    out, ob, ro, float(reg), _stream())
         else:
             # slab the normal equations: A is nrows*k*k fp32, which at the
             # 1B-rating configs would exceed HBM if materialized whole.
@@ -114,35 +154,57 @@ def als_solve_side(
             A = torch.empty(slab, k, k, dtype=torch.float32,
                             device=fac.device)
             b = torch.empty(slab, k, dtype=torch.float32, device=fac.device)
+            # degree-descending schedule: only meaningful when one slab
+            # covers the side (the gramian then scatters A/b by entity and
+            # the solve stays order-agnostic); multi-slab runs skip it
+            use_ro = ro if slab >= csr.num_rows else _empty(fac.device)
             for s in range(0, csr.num_rows, slab):
                 e = min(s + slab, csr.num_rows)
                 As, bs = A[: e - s], b[: e - s]
-                ops.gramian(csr.indptr[s:e + 1], csr.indices, csr.values,
-                            fac, As, bs, float(reg), _stream())
+                gram = ops.gramian_fp8 if fp8 else ops.gramian
+                gram(csr.indptr[s:e + 1], csr.indices, csr.values,
+                     fac, As, bs, use_ro, float(reg), _stream())
                 obs = ob[s:e] if ob.numel() > 0 else ob
+                o8s = o8[s:e] if o8.numel() > 0 else o8
                 if k <= 64:
-                    ops.ldl_solve_wave_reg(As, bs, out[s:e], obs, _stream())
+                    ops.ldl_solve_wave_reg(As, bs, out[s:e], obs, o8s,
+                                           _stream())
                 else:
                     ops.cholesky_solve(As, bs, out[s:e], _stream())
                     if out_bf16 is not None:
                         out_bf16[s:e].copy_(out[s:e].to(torch.bfloat16))
+                    if out_fp8 is not None:
+                        out_fp8[s:e].copy_(quantize_fp8(out[s:e]))
         return out[:, :korig] if korig != k else out
+    if other_factors.dtype == torch.uint8:   # CPU fp8 emulation
+        out = reference.als_solve_side_reference(
+            csr, dequantize_fp8(other_factors), reg)
+        if out_fp8 is not None:
+            out_fp8.copy_(quantize_fp8(out[:, : out_fp8.shape[1]]))
+        return out
     return reference.als_solve_side_reference(csr, other_factors, reg)
 
 
 def gramian(csr: CSR, factors: torch.Tensor, reg: float) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Standalone K1 (parity tests / modular path).  ``factors`` as uint8 =
+    e4m3 bytes -> the fp8 gather kernel; else bf16."""
     if factors.is_cuda:
         ops = _require_hip()
-        fac = _pad_k(factors.to(torch.bfloat16).contiguous())
+        fp8 = factors.dtype == torch.uint8
+        fac = (factors.contiguous() if fp8
+               else _pad_k(factors.to(torch.bfloat16).contiguous()))
         k = fac.shape[1]
         A = torch.empty(csr.num_rows, k, k, dtype=torch.float32, device=fac.device)
         b = torch.empty(csr.num_rows, k, dtype=torch.float32, device=fac.device)
-        ops.gramian(csr.indptr, csr.indices, csr.values, fac, A, b,
-                    float(reg), _stream())
+        fn = ops.gramian_fp8 if fp8 else ops.gramian
+        fn(csr.indptr, csr.indices, csr.values, fac, A, b,
+           _empty(fac.device), float(reg), _stream())
         korig = factors.shape[1]
         if korig != k:
             return A[:, :korig, :korig], b[:, :korig]
         return A, b
+    if factors.dtype == torch.uint8:
+        return reference.gramian_reference(csr, dequantize_fp8(factors), reg)
     return reference.gramian_reference(csr, factors, reg)
 
 

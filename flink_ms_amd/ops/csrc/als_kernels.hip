@@ -69,6 +69,10 @@ __global__ void k_als_solve_fused(const long long* __restrict__ indptr,
 }
 
 // Standalone K1 (for parity tests / modular path): writes dense A and b.
+// row_order (optional): degree-descending launch schedule — block i
+// processes entity row_order[i] and writes A_out/b_out at THAT row, so the
+// downstream batched solve stays order-agnostic (heavy entities launch
+// first; avoids tail stragglers at the end of the grid).
 template <int KT>
 __launch_bounds__(256)
 __global__ void k_gramian(const long long* __restrict__ indptr,
@@ -77,15 +81,86 @@ __global__ void k_gramian(const long long* __restrict__ indptr,
                           const unsigned short* __restrict__ factors,
                           float* __restrict__ A_out,   // [nrows][K][K]
                           float* __restrict__ b_out,   // [nrows][K]
+                          const int* __restrict__ row_order,
                           long long nrows, float reg) {
     constexpr int K = Geo<KT>::K;
     __shared__ __align__(16) char smem[Geo<KT>::SMEM];
-    const long long row = blockIdx.x;
+    long long row = blockIdx.x;
     if (row >= nrows) return;
+    if (row_order) row = row_order[row];
     const int tid = threadIdx.x;
 
     const int n = gramian_to_lds<KT>(smem, indptr, indices, values, factors,
                                      row, reg);
+    float* A = (float*)smem;
+    float* b = A + K * (K + 1);
+    if (n == 0) {
+        for (int i = tid; i < K * K; i += 256) A_out[row * K * K + i] = 0.0f;
+        for (int c = tid; c < K; c += 256) b_out[row * K + c] = 0.0f;
+        return;
+    }
+    for (int i = tid; i < K * K; i += 256)
+        A_out[row * K * K + i] = A[(i / K) * (K + 1) + (i % K)];
+    for (int c = tid; c < K; c += 256) b_out[row * K + c] = b[c];
+}
+
+// fp8 twins of the fused and standalone-K1 kernels: e4m3 factor gathers
+// (one cache line per k<=64 row) through gramian_to_lds_fp8; everything
+// downstream (fp32 A/b, LDL solve) is identical.  out_fp8 writes the next
+// half-iteration's e4m3 factor image alongside fp32 (and optional bf16).
+template <int KT>
+__launch_bounds__(256)
+__global__ void k_als_solve_fused_fp8(const long long* __restrict__ indptr,
+                                      const int* __restrict__ indices,
+                                      const float* __restrict__ values,
+                                      const unsigned char* __restrict__ factors,
+                                      float* __restrict__ out_f32,
+                                      unsigned char* __restrict__ out_fp8,
+                                      const int* __restrict__ row_order,
+                                      long long nrows, float reg) {
+    constexpr int K = Geo<KT>::K;
+    __shared__ __align__(16) char smem[Geo<KT>::SMEM_TRI];
+    long long row = blockIdx.x;
+    if (row >= nrows) return;
+    if (row_order) row = row_order[row];
+
+    const int n = gramian_to_lds_fp8<KT, true, true>(smem, indptr, indices,
+                                                     values, factors, row,
+                                                     reg);
+    if (n == 0) {
+        for (int c = threadIdx.x; c < K; c += 256) {
+            out_f32[row * K + c] = 0.0f;
+            if (out_fp8) out_fp8[row * K + c] = 0;
+        }
+        return;
+    }
+    float* A = (float*)smem;
+    float* b = A + Geo<KT>::A_TRI_FLOATS;
+    cholesky_lds_tri<K>(A, b + 2 * K);
+    solve_lds_block_tri<K>(A, b, b + 2 * K, out_f32 + row * K, nullptr,
+                           out_fp8 ? out_fp8 + row * K : nullptr);
+}
+
+template <int KT>
+__launch_bounds__(256)
+__global__ void k_gramian_fp8(const long long* __restrict__ indptr,
+                              const int* __restrict__ indices,
+                              const float* __restrict__ values,
+                              const unsigned char* __restrict__ factors,
+                              float* __restrict__ A_out,   // [nrows][K][K]
+                              float* __restrict__ b_out,   // [nrows][K]
+                              const int* __restrict__ row_order,
+                              long long nrows, float reg) {
+    constexpr int K = Geo<KT>::K;
+    static_assert(Geo<KT>::STAGE8_BYTES <= Geo<KT>::SMEM, "stage fits");
+    __shared__ __align__(16) char smem[Geo<KT>::SMEM];
+    long long row = blockIdx.x;
+    if (row >= nrows) return;
+    if (row_order) row = row_order[row];
+    const int tid = threadIdx.x;
+
+    const int n = gramian_to_lds_fp8<KT>(smem, indptr, indices, values,
+                                         factors, row, reg);
     float* A = (float*)smem;
     float* b = A + K * (K + 1);
     if (n == 0) {
@@ -192,6 +267,32 @@ __global__ void k_mfma_probe_bf16(const unsigned short* __restrict__ Xt,
         C[(g * 4 + r) * 16 + li] = acc[r];
 }
 
+// fp8 e4m3 16x16x32 probe through the fp8 stage geometry (TROW8 rows,
+// ds_read_b64 fragments): C[16][16] = Xt^T @ Yt for Xt, Yt [32][16] e4m3
+// bytes.  Validates the (lane, element) map + C/D layout assumption of
+// k_gramian_fp8 against a torch reference (tests/test_gpu_mfma.py).
+__global__ void k_mfma_probe_fp8(const unsigned char* __restrict__ Xt,
+                                 const unsigned char* __restrict__ Yt,
+                                 float* __restrict__ C) {
+    constexpr int TROW8 = 40;
+    __shared__ __align__(8) char st[32 * TROW8];
+    const int lane = threadIdx.x;  // launched with 64 threads
+    for (int i = lane; i < 2 * 16 * 32; i += 64) {
+        const int tile = i >> 9, rem = i & 511;
+        const int r = rem >> 4, c = rem & 15;   // rating r, column c
+        st[(tile * 16 + c) * TROW8 + r] = (tile == 0 ? Xt : Yt)[r * 16 + c];
+    }
+    __syncthreads();
+    const int g = lane >> 4, li = lane & 15;
+    const fp8x8 fx = *(const fp8x8*)(st + (li)*TROW8 + g * 8);
+    const fp8x8 fy = *(const fp8x8*)(st + (16 + li) * TROW8 + g * 8);
+    f32x4 acc{0, 0, 0, 0};
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(fx, fy, acc, 0, 0, 0);
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+        C[(g * 4 + r) * 16 + li] = acc[r];
+}
+
 // -------------------------------------------------------------- launchers
 
 extern "C" hipError_t fma_als_solve_fused(
@@ -209,11 +310,36 @@ extern "C" hipError_t fma_als_solve_fused(
 extern "C" hipError_t fma_gramian(
     int k, const long long* indptr, const int* indices, const float* values,
     const unsigned short* factors, float* A_out, float* b_out,
-    long long nrows, float reg, hipStream_t stream) {
+    const int* row_order, long long nrows, float reg, hipStream_t stream) {
     if (k % 16 || k < 16 || k > 128 || nrows <= 0) return hipErrorInvalidValue;
     dim3 grid((unsigned)nrows), block(256);
     DISPATCH_KT(k, (k_gramian<KT><<<grid, block, 0, stream>>>(
-        indptr, indices, values, factors, A_out, b_out, nrows, reg)));
+        indptr, indices, values, factors, A_out, b_out, row_order, nrows,
+        reg)));
+    return hipGetLastError();
+}
+
+extern "C" hipError_t fma_gramian_fp8(
+    int k, const long long* indptr, const int* indices, const float* values,
+    const unsigned char* factors, float* A_out, float* b_out,
+    const int* row_order, long long nrows, float reg, hipStream_t stream) {
+    if (k % 16 || k < 16 || k > 128 || nrows <= 0) return hipErrorInvalidValue;
+    dim3 grid((unsigned)nrows), block(256);
+    DISPATCH_KT(k, (k_gramian_fp8<KT><<<grid, block, 0, stream>>>(
+        indptr, indices, values, factors, A_out, b_out, row_order, nrows,
+        reg)));
+    return hipGetLastError();
+}
+
+extern "C" hipError_t fma_als_solve_fused_fp8(
+    int k, const long long* indptr, const int* indices, const float* values,
+    const unsigned char* factors, float* out_f32, unsigned char* out_fp8,
+    const int* row_order, long long nrows, float reg, hipStream_t stream) {
+    if (k % 16 || k < 16 || k > 128 || nrows <= 0) return hipErrorInvalidValue;
+    dim3 grid((unsigned)nrows), block(256);
+    DISPATCH_KT(k, (k_als_solve_fused_fp8<KT><<<grid, block, 0, stream>>>(
+        indptr, indices, values, factors, out_f32, out_fp8, row_order,
+        nrows, reg)));
     return hipGetLastError();
 }
 
@@ -245,6 +371,7 @@ __global__ void k_ldl_solve_wave(const float* __restrict__ A_in,
                                  const float* __restrict__ b_in,
                                  float* __restrict__ x_out,
                                  unsigned short* __restrict__ x_bf16,
+                                 unsigned char* __restrict__ x_fp8,
                                  long long nrows) {
     constexpr int K = KT * 16;
     static_assert(K <= 64, "wave solver handles k <= 64");
@@ -389,6 +516,7 @@ __global__ void k_ldl_solve_wave(const float* __restrict__ A_in,
     if (lane < K) {
         x_out[e * K + lane] = x0;
         if (x_bf16) x_bf16[e * K + lane] = f2bf(x0);
+        if (x_fp8) x_fp8[e * K + lane] = f2fp8(x0);
     }
 }
 
@@ -572,6 +700,7 @@ __global__ void k_ldl_solve_wave_reg(const float* __restrict__ A_in,
                                      const float* __restrict__ b_in,
                                      float* __restrict__ x_out,
                                      unsigned short* __restrict__ x_bf16,
+                                     unsigned char* __restrict__ x_fp8,
                                      long long nrows) {
     constexpr int K = KT * 16;
     static_assert(K <= 64, "register wave solver handles k <= 64");
@@ -597,19 +726,21 @@ __global__ void k_ldl_solve_wave_reg(const float* __restrict__ A_in,
     if (lane < K) {
         x_out[e * K + lane] = x0;
         if (x_bf16) x_bf16[e * K + lane] = f2bf(x0);
+        if (x_fp8) x_fp8[e * K + lane] = f2fp8(x0);
     }
 }
 
 extern "C" hipError_t fma_ldl_solve_wave_reg(
     int k, const float* A_in, const float* b_in, float* x_out,
-    unsigned short* x_bf16, long long nrows, hipStream_t stream) {
+    unsigned short* x_bf16, unsigned char* x_fp8, long long nrows,
+    hipStream_t stream) {
     if (k % 16 || k < 16 || k > 64 || nrows <= 0) return hipErrorInvalidValue;
     dim3 grid((unsigned)((nrows + 3) / 4)), block(256);
     switch (k / 16) {
-        case 1: k_ldl_solve_wave_reg<1><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, nrows); break;
-        case 2: k_ldl_solve_wave_reg<2><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, nrows); break;
-        case 3: k_ldl_solve_wave_reg<3><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, nrows); break;
-        case 4: k_ldl_solve_wave_reg<4><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, nrows); break;
+        case 1: k_ldl_solve_wave_reg<1><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, x_fp8, nrows); break;
+        case 2: k_ldl_solve_wave_reg<2><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, x_fp8, nrows); break;
+        case 3: k_ldl_solve_wave_reg<3><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, x_fp8, nrows); break;
+        case 4: k_ldl_solve_wave_reg<4><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, x_fp8, nrows); break;
         default: return hipErrorInvalidValue;
     }
     return hipGetLastError();
@@ -617,14 +748,15 @@ extern "C" hipError_t fma_ldl_solve_wave_reg(
 
 extern "C" hipError_t fma_ldl_solve_wave(
     int k, const float* A_in, const float* b_in, float* x_out,
-    unsigned short* x_bf16, long long nrows, hipStream_t stream) {
+    unsigned short* x_bf16, unsigned char* x_fp8, long long nrows,
+    hipStream_t stream) {
     if (k % 16 || k < 16 || k > 64 || nrows <= 0) return hipErrorInvalidValue;
     dim3 grid((unsigned)((nrows + 3) / 4)), block(256);
     switch (k / 16) {
-        case 1: k_ldl_solve_wave<1><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, nrows); break;
-        case 2: k_ldl_solve_wave<2><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, nrows); break;
-        case 3: k_ldl_solve_wave<3><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, nrows); break;
-        case 4: k_ldl_solve_wave<4><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, nrows); break;
+        case 1: k_ldl_solve_wave<1><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, x_fp8, nrows); break;
+        case 2: k_ldl_solve_wave<2><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, x_fp8, nrows); break;
+        case 3: k_ldl_solve_wave<3><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, x_fp8, nrows); break;
+        case 4: k_ldl_solve_wave<4><<<grid, block, 0, stream>>>(A_in, b_in, x_out, x_bf16, x_fp8, nrows); break;
         default: return hipErrorInvalidValue;
     }
     return hipGetLastError();
@@ -640,6 +772,13 @@ extern "C" hipError_t fma_mfma_probe_bf16(const unsigned short* Xt,
                                           const unsigned short* Yt, float* C,
                                           hipStream_t stream) {
     k_mfma_probe_bf16<<<dim3(1), dim3(64), 0, stream>>>(Xt, Yt, C);
+    return hipGetLastError();
+}
+
+extern "C" hipError_t fma_mfma_probe_fp8(const unsigned char* Xt,
+                                         const unsigned char* Yt, float* C,
+                                         hipStream_t stream) {
+    k_mfma_probe_fp8<<<dim3(1), dim3(64), 0, stream>>>(Xt, Yt, C);
     return hipGetLastError();
 }
 

@@ -18,8 +18,17 @@ int fma_als_solve_fused(int k, const int64_t* indptr, const int* indices,
                         void* stream);
 int fma_gramian(int k, const int64_t* indptr, const int* indices,
                 const float* values, const unsigned short* factors,
-                float* A_out, float* b_out, int64_t nrows, float reg,
-                void* stream);
+                float* A_out, float* b_out, const int* row_order,
+                int64_t nrows, float reg, void* stream);
+int fma_gramian_fp8(int k, const int64_t* indptr, const int* indices,
+                    const float* values, const unsigned char* factors,
+                    float* A_out, float* b_out, const int* row_order,
+                    int64_t nrows, float reg, void* stream);
+int fma_als_solve_fused_fp8(int k, const int64_t* indptr, const int* indices,
+                            const float* values, const unsigned char* factors,
+                            float* out_f32, unsigned char* out_fp8,
+                            const int* row_order, int64_t nrows, float reg,
+                            void* stream);
 int fma_cholesky_solve(int k, const float* A_in, const float* b_in,
                        float* x_out, int64_t nrows, void* stream);
 int fma_cholesky_solve_ph(int k, const float* A_in, const float* b_in,
@@ -40,14 +49,17 @@ int fma_sgd_update(unsigned short* U, unsigned short* V,
                    const float* r, float* err_out, int64_t nq, int k,
                    float lr, float user_reg, float item_reg, void* stream);
 int fma_ldl_solve_wave(int k, const float* A_in, const float* b_in,
-                       float* x_out, unsigned short* x_bf16, int64_t nrows,
-                       void* stream);
+                       float* x_out, unsigned short* x_bf16,
+                       unsigned char* x_fp8, int64_t nrows, void* stream);
 int fma_ldl_solve_wave_reg(int k, const float* A_in, const float* b_in,
                            float* x_out, unsigned short* x_bf16,
-                           int64_t nrows, void* stream);
+                           unsigned char* x_fp8, int64_t nrows,
+                           void* stream);
 int fma_mfma_probe_f32(const float* A, const float* B, float* D, void* stream);
 int fma_mfma_probe_bf16(const unsigned short* Xt, const unsigned short* Yt,
                         float* C, void* stream);
+int fma_mfma_probe_fp8(const unsigned char* Xt, const unsigned char* Yt,
+                       float* C, void* stream);
 int fma_dbg_stage_dump(int k, const int64_t* indptr, const int* indices,
                        const float* values, const unsigned short* factors,
                        unsigned short* out, void* stream);
@@ -82,6 +94,27 @@ unsigned short* bf16_ptr_mut(torch::Tensor& t) {
     TORCH_CHECK(t.scalar_type() == torch::kBFloat16,
                 "expected a bf16 output tensor at the kernel boundary");
     return reinterpret_cast<unsigned short*>(t.data_ptr());
+}
+
+// e4m3 factor images travel as uint8 byte tensors
+const unsigned char* fp8_ptr(const torch::Tensor& t) {
+    TORCH_CHECK(t.scalar_type() == torch::kUInt8,
+                "expected a uint8 (e4m3) tensor at the kernel boundary");
+    return reinterpret_cast<const unsigned char*>(t.data_ptr());
+}
+unsigned char* fp8_ptr_mut(torch::Tensor& t) {
+    TORCH_CHECK(t.scalar_type() == torch::kUInt8,
+                "expected a uint8 (e4m3) output tensor at the kernel boundary");
+    return reinterpret_cast<unsigned char*>(t.data_ptr());
+}
+
+const int* order_ptr(const torch::Tensor& row_order, int64_t nrows) {
+    if (row_order.numel() == 0) return nullptr;
+    TORCH_CHECK(row_order.scalar_type() == torch::kInt32 &&
+                row_order.is_cuda() && row_order.is_contiguous(),
+                "row_order must be contiguous int32 on GPU");
+    TORCH_CHECK(row_order.numel() == nrows, "row_order size");
+    return row_order.data_ptr<int>();
 }
 
 void als_solve_fused(torch::Tensor indptr, torch::Tensor indices,
@@ -119,7 +152,7 @@ void als_solve_fused(torch::Tensor indptr, torch::Tensor indices,
 
 void gramian(torch::Tensor indptr, torch::Tensor indices, torch::Tensor values,
              torch::Tensor factors, torch::Tensor A_out, torch::Tensor b_out,
-             double reg, int64_t stream) {
+             torch::Tensor row_order, double reg, int64_t stream) {
     check_t(indptr, torch::kInt64, "indptr");
     check_t(indices, torch::kInt32, "indices");
     check_t(values, torch::kFloat32, "values");
@@ -131,9 +164,59 @@ void gramian(torch::Tensor indptr, torch::Tensor indices, torch::Tensor values,
     check_hip(fma_gramian(k, indptr.data_ptr<int64_t>(),
                           indices.data_ptr<int>(), values.data_ptr<float>(),
                           bf16_ptr(factors), A_out.data_ptr<float>(),
-                          b_out.data_ptr<float>(), nrows, (float)reg,
+                          b_out.data_ptr<float>(),
+                          order_ptr(row_order, nrows), nrows, (float)reg,
                           (void*)stream),
               "gramian");
+}
+
+void gramian_fp8(torch::Tensor indptr, torch::Tensor indices,
+                 torch::Tensor values, torch::Tensor factors,
+                 torch::Tensor A_out, torch::Tensor b_out,
+                 torch::Tensor row_order, double reg, int64_t stream) {
+    check_t(indptr, torch::kInt64, "indptr");
+    check_t(indices, torch::kInt32, "indices");
+    check_t(values, torch::kFloat32, "values");
+    check_t(factors, torch::kUInt8, "factors");
+    check_t(A_out, torch::kFloat32, "A_out");
+    check_t(b_out, torch::kFloat32, "b_out");
+    const int k = (int)factors.size(1);
+    const long long nrows = indptr.size(0) - 1;
+    check_hip(fma_gramian_fp8(
+                  k, indptr.data_ptr<int64_t>(), indices.data_ptr<int>(),
+                  values.data_ptr<float>(), fp8_ptr(factors),
+                  A_out.data_ptr<float>(), b_out.data_ptr<float>(),
+                  order_ptr(row_order, nrows), nrows, (float)reg,
+                  (void*)stream),
+              "gramian_fp8");
+}
+
+void als_solve_fused_fp8(torch::Tensor indptr, torch::Tensor indices,
+                         torch::Tensor values, torch::Tensor factors,
+                         torch::Tensor out_f32, torch::Tensor out_fp8,
+                         torch::Tensor row_order, double reg,
+                         int64_t stream) {
+    check_t(indptr, torch::kInt64, "indptr");
+    check_t(indices, torch::kInt32, "indices");
+    check_t(values, torch::kFloat32, "values");
+    check_t(factors, torch::kUInt8, "factors");
+    check_t(out_f32, torch::kFloat32, "out_f32");
+    const int k = (int)factors.size(1);
+    const long long nrows = indptr.size(0) - 1;
+    TORCH_CHECK(out_f32.size(0) == nrows && out_f32.size(1) == k,
+                "out_f32 shape mismatch");
+    unsigned char* o8 = nullptr;
+    if (out_fp8.numel() > 0) {
+        check_t(out_fp8, torch::kUInt8, "out_fp8");
+        TORCH_CHECK(out_fp8.numel() == out_f32.numel(), "out_fp8 shape");
+        o8 = fp8_ptr_mut(out_fp8);
+    }
+    check_hip(fma_als_solve_fused_fp8(
+                  k, indptr.data_ptr<int64_t>(), indices.data_ptr<int>(),
+                  values.data_ptr<float>(), fp8_ptr(factors),
+                  out_f32.data_ptr<float>(), o8, order_ptr(row_order, nrows),
+                  nrows, (float)reg, (void*)stream),
+              "als_solve_fused_fp8");
 }
 
 void cholesky_solve_ph(torch::Tensor A, torch::Tensor b, torch::Tensor x,
@@ -145,21 +228,25 @@ void cholesky_solve_ph(torch::Tensor A, torch::Tensor b, torch::Tensor x,
 }
 
 void ldl_solve_wave(torch::Tensor A, torch::Tensor b, torch::Tensor x,
-                    torch::Tensor x_bf16, int64_t stream) {
+                    torch::Tensor x_bf16, torch::Tensor x_fp8,
+                    int64_t stream) {
     unsigned short* xb = x_bf16.numel() > 0 ? bf16_ptr_mut(x_bf16) : nullptr;
+    unsigned char* x8 = x_fp8.numel() > 0 ? fp8_ptr_mut(x_fp8) : nullptr;
     check_hip(fma_ldl_solve_wave((int)A.size(1), A.data_ptr<float>(),
                                  b.data_ptr<float>(), x.data_ptr<float>(),
-                                 xb, A.size(0), (void*)stream),
+                                 xb, x8, A.size(0), (void*)stream),
               "ldl_solve_wave");
 }
 
 // register-resident v2 (A in MFMA fragments, panel scratch in LDS)
 void ldl_solve_wave_reg(torch::Tensor A, torch::Tensor b, torch::Tensor x,
-                        torch::Tensor x_bf16, int64_t stream) {
+                        torch::Tensor x_bf16, torch::Tensor x_fp8,
+                        int64_t stream) {
     unsigned short* xb = x_bf16.numel() > 0 ? bf16_ptr_mut(x_bf16) : nullptr;
+    unsigned char* x8 = x_fp8.numel() > 0 ? fp8_ptr_mut(x_fp8) : nullptr;
     check_hip(fma_ldl_solve_wave_reg((int)A.size(1), A.data_ptr<float>(),
                                      b.data_ptr<float>(), x.data_ptr<float>(),
-                                     xb, A.size(0), (void*)stream),
+                                     xb, x8, A.size(0), (void*)stream),
               "ldl_solve_wave_reg");
 }
 
@@ -272,6 +359,16 @@ void mfma_probe_bf16(torch::Tensor Xt, torch::Tensor Yt, torch::Tensor C,
               "mfma_probe_bf16");
 }
 
+void mfma_probe_fp8(torch::Tensor Xt, torch::Tensor Yt, torch::Tensor C,
+                    int64_t stream) {
+    check_t(Xt, torch::kUInt8, "Xt");
+    check_t(Yt, torch::kUInt8, "Yt");
+    check_t(C, torch::kFloat32, "C");
+    check_hip(fma_mfma_probe_fp8(fp8_ptr(Xt), fp8_ptr(Yt),
+                                 C.data_ptr<float>(), (void*)stream),
+              "mfma_probe_fp8");
+}
+
 void dbg_stage_dump(torch::Tensor indptr, torch::Tensor indices,
                     torch::Tensor values, torch::Tensor factors,
                     torch::Tensor out, int64_t stream) {
@@ -314,6 +411,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "flink_ms_amd MI355X (gfx950) HIP kernels";
     m.def("als_solve_fused", &als_solve_fused);
     m.def("gramian", &gramian);
+    m.def("gramian_fp8", &gramian_fp8);
+    m.def("als_solve_fused_fp8", &als_solve_fused_fp8);
     m.def("cholesky_solve", &cholesky_solve);
     m.def("cholesky_solve_ph", &cholesky_solve_ph);
     m.def("ldl_solve_wave", &ldl_solve_wave);
@@ -327,4 +426,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("dbg_gramian", &dbg_gramian);
     m.def("mfma_probe_f32", &mfma_probe_f32);
     m.def("mfma_probe_bf16", &mfma_probe_bf16);
+    m.def("mfma_probe_fp8", &mfma_probe_fp8);
 }

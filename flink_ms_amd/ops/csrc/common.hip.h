@@ -2,6 +2,7 @@
 // Single HIP path, MI355X-only: wave64, MFMA bf16 16x16x32 tiles, LDS staging.
 #pragma once
 
+#include <hip/hip_fp8.h>
 #include <hip/hip_runtime.h>
 
 #define WAVE 64
@@ -9,8 +10,25 @@
 
 // MFMA fragment/accumulator register types (cdna_hip_programming.md §3):
 // 8 bf16 (4 VGPRs) per A/B fragment, 4 fp32 accumulators per 16x16 C/D tile.
+// fp8 e4m3 fragments are 8 bytes (2 VGPRs): the i64 operand of
+// v_mfma_f32_16x16x32_fp8_fp8 (OCP e4m3fn on gfx950, NOT MI300X fnuz).
 typedef short bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
+typedef long long fp8x8;
+
+// float <-> OCP e4m3fn scalar conversions (round-to-nearest-even, matching
+// torch.float8_e4m3fn casts; conversions sit in staging/epilogue, not the
+// MFMA hot loop).
+DEV_INLINE unsigned char f2fp8(float f) {
+    __hip_fp8_e4m3 q(f);
+    return q.__x;
+}
+
+DEV_INLINE float fp82f(unsigned char u) {
+    __hip_fp8_e4m3 q;
+    q.__x = u;
+    return (float)q;
+}
 
 DEV_INLINE float bf2f(unsigned short u) {
     union { unsigned int i; float f; } v;

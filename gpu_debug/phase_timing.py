@@ -1,3 +1,5 @@
+import torch as _t
+_e8 = _t.empty(0)  # empty sentinel: optional tensor args (row_order/x_fp8)
 """Split the fused ALS solve cost: chunk loop (K1) vs Cholesky+solve (K2).
 
 Times, on the real ML-25M user-side problem (162K entities, 25M ratings,
@@ -46,7 +48,7 @@ x = torch.empty(csr.num_rows, k, dtype=torch.float32, device=dev)
 timeit("user fused (gramian+chol)", lambda: hip.als_solve_fused(
     csr.indptr, csr.indices, csr.values, V, out, emptyb, emptyi, 0.9, st()))
 timeit("user gramian only", lambda: hip.gramian(
-    csr.indptr, csr.indices, csr.values, V, A, b, 0.9, st()))
+    csr.indptr, csr.indices, csr.values, V, A, b, _e8, 0.9, st()))
 timeit("user cholesky only", lambda: hip.cholesky_solve(A, b, x, st()))
 
 outi = torch.empty(icsr.num_rows, k, dtype=torch.float32, device=dev)
@@ -55,7 +57,7 @@ bi = torch.empty(icsr.num_rows, k, dtype=torch.float32, device=dev)
 timeit("item fused", lambda: hip.als_solve_fused(
     icsr.indptr, icsr.indices, icsr.values, U, outi, emptyb, emptyi, 0.9, st()))
 timeit("item gramian only", lambda: hip.gramian(
-    icsr.indptr, icsr.indices, icsr.values, U, Ai, bi, 0.9, st()))
+    icsr.indptr, icsr.indices, icsr.values, U, Ai, bi, _e8, 0.9, st()))
 
 # row_order effect
 order = torch.argsort(csr.row_counts(), descending=True).to(torch.int32).to(dev)
@@ -69,12 +71,12 @@ timeit("chol solve-only (2)", lambda: hip.cholesky_solve_ph(A, b, x, 2, st()))
 timeit("chol full (3)", lambda: hip.cholesky_solve_ph(A, b, x, 3, st()))
 
 xb = torch.empty(csr.num_rows, k, dtype=torch.bfloat16, device=dev)
-timeit("ldl wave solver (k=64)", lambda: hip.ldl_solve_wave(A, b, x, xb, st()))
+timeit("ldl wave solver (k=64)", lambda: hip.ldl_solve_wave(A, b, x, xb, _e8, st()))
 def modular_user():
-    hip.gramian(csr.indptr, csr.indices, csr.values, V, A, b, 0.9, st())
-    hip.ldl_solve_wave(A, b, x, xb, st())
+    hip.gramian(csr.indptr, csr.indices, csr.values, V, A, b, _e8, 0.9, st())
+    hip.ldl_solve_wave(A, b, x, xb, _e8, st())
 timeit("user modular gramian+wave-solve", modular_user)
 def modular_item():
-    hip.gramian(icsr.indptr, icsr.indices, icsr.values, U, Ai, bi, 0.9, st())
-    xi = outi; hip.ldl_solve_wave(Ai, bi, xi, emptyb, st())
+    hip.gramian(icsr.indptr, icsr.indices, icsr.values, U, Ai, bi, _e8, 0.9, st())
+    xi = outi; hip.ldl_solve_wave(Ai, bi, xi, emptyb, _e8, st())
 timeit("item modular gramian+wave-solve", modular_item)

@@ -1,3 +1,5 @@
+import torch as _t
+_e8 = _t.empty(0)  # empty sentinel: optional tensor args (row_order/x_fp8)
 """k=128 cost decomposition on the 1B-config per-GPU shape."""
 
 import torch
@@ -35,8 +37,7 @@ b = torch.empty(SLAB, k, dtype=torch.float32, device=dev)
 x = torch.empty(SLAB, k, dtype=torch.float32, device=dev)
 
 def gram_slab():
-    hip.gramian(csr.indptr[:SLAB + 1], csr.indices, csr.values, V, A, b,
-                0.9, st())
+    hip.gramian(csr.indptr[:SLAB + 1], csr.indices, csr.values, V, A, b, _e8, 0.9, st())
 timeit(f"user gramian slab ({SLAB} rows, k=128)", gram_slab)
 timeit("chol load-only (0)", lambda: hip.cholesky_solve_ph(A, b, x, 0, st()))
 timeit("chol eliminate (1)", lambda: hip.cholesky_solve_ph(A, b, x, 1, st()))

@@ -1,3 +1,5 @@
+import torch as _t
+_e8 = _t.empty(0)  # empty sentinel: optional tensor args (row_order/x_fp8)
 """A/B: register-resident wave LDL (v2) vs triangular-LDS wave LDL (v1).
 
 Parity vs the fp32 torch reference first (k in 16/32/48/64), then timing
@@ -57,11 +59,10 @@ def timeit(name, fn, reps=10):
 for nm, c, fac in (("user", csr, V), ("item", icsr, U)):
     A = torch.empty(c.num_rows, k, k, dtype=torch.float32, device=dev)
     b = torch.empty(c.num_rows, k, dtype=torch.float32, device=dev)
-    hip.gramian(c.indptr, c.indices, c.values, fac, A, b, 0.9, st())
+    hip.gramian(c.indptr, c.indices, c.values, fac, A, b, _e8, 0.9, st())
     x = torch.empty(c.num_rows, k, dtype=torch.float32, device=dev)
     xb = torch.empty(c.num_rows, k, dtype=torch.bfloat16, device=dev)
     timeit(f"{nm} solve v1 (tri-LDS)",
-           lambda A=A, b=b, x=x, xb=xb: hip.ldl_solve_wave(A, b, x, xb, st()))
+           lambda A=A, b=b, x=x, xb=xb: hip.ldl_solve_wave(A, b, x, xb, _e8, st()))
     timeit(f"{nm} solve v2 (register)",
-           lambda A=A, b=b, x=x, xb=xb: hip.ldl_solve_wave_reg(A, b, x, xb,
-                                                               st()))
+           lambda A=A, b=b, x=x, xb=xb: hip.ldl_solve_wave_reg(A, b, x, xb, _e8, st()))

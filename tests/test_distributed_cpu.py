@@ -288,3 +288,51 @@ def test_svm_alpha_averaging_in_place():
         assert res[rank]["moved"], "SDCA made no dual progress"
         assert res[rank]["same_buffer"], "alpha was rebound (graph unsafe)"
         assert res[rank]["averaged"], "CoCoA 1/K dual averaging not applied"
+
+
+def _als_fp8_worker(rank, world, port, q):
+    """fp8 factor exchange at world 2: uint8 e4m3 shards ride the gloo
+    (RCCL on GPU) all-gather — half the bf16 wire bytes — and distributed
+    training quality must match the single-process fp8 run."""
+    torch.manual_seed(0)
+    ctx = _init(rank, world, port)
+    from flink_ms_amd.data.ratings import RatingsShape, synthetic_ratings
+    from flink_ms_amd.models.als import ALSConfig, ALSTrainer
+    shape = RatingsShape(120, 60, 2000)
+    u, i, r = synthetic_ratings(shape, seed=5)
+    mask = torch.arange(shape.num_ratings) % world == rank
+    cfg = ALSConfig(iterations=3, num_factors=8, lambda_=0.2,
+                    dtype=torch.float32, factor_dtype="fp8")
+    tr = ALSTrainer(cfg, ctx)
+    tr.setup(u[mask].long(), i[mask].long(), r[mask],
+             shape.num_users, shape.num_items)
+    assert tr.item_shard.dtype == torch.uint8
+    tr.fit()
+    m = tr.model()
+    q.put((rank, {"uf": m.user_factors.tolist(),
+                  "if": m.item_factors.tolist()}))
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_distributed_als_fp8_exchange():
+    results = _run_workers(_als_fp8_worker)
+    import flink_ms_amd.parallel.dist as D
+    D._CTX = None
+    for v in ("RANK", "WORLD_SIZE", "MASTER_ADDR", "MASTER_PORT"):
+        os.environ.pop(v, None)
+    from flink_ms_amd.data.ratings import RatingsShape, synthetic_ratings
+    from flink_ms_amd.models.als import ALSConfig, train_als
+    from flink_ms_amd.models.mse import evaluate_mse
+    shape = RatingsShape(120, 60, 2000)
+    u, i, r = synthetic_ratings(shape, seed=5)
+    uf = torch.tensor(results[0]["uf"] + results[1]["uf"])
+    itf = torch.tensor(results[0]["if"] + results[1]["if"])
+    res = evaluate_mse(uf, itf, u, i, r)
+    model_sp, _ = train_als(
+        u.long(), i.long(), r, shape.num_users, shape.num_items,
+        ALSConfig(iterations=3, num_factors=8, lambda_=0.2,
+                  dtype=torch.float32, factor_dtype="fp8"))
+    res_sp = evaluate_mse(model_sp.user_factors, model_sp.item_factors,
+                          u, i, r)
+    assert res.mse < res_sp.mse * 1.25 + 0.05, (res.mse, res_sp.mse)

@@ -37,3 +37,21 @@ def test_mfma_bf16_probe_gramian_path(gpu):
     torch.cuda.synchronize()
     ref = Xt.to(torch.float32).T @ Yt.to(torch.float32)
     assert torch.allclose(C, ref, atol=1e-2), (C - ref).abs().max()
+
+
+def test_mfma_probe_fp8(gpu):
+    """fp8 e4m3 16x16x32 through the fp8 stage geometry: C = Xt^T @ Yt.
+    Asymmetric operands (transpose-detecting, guide §5.4 rule 16)."""
+    import flink_ms_amd._hip_ops as hip
+    from flink_ms_amd import ops
+    g = torch.Generator().manual_seed(77)
+    X = torch.randn(32, 16, generator=g) * 0.5
+    Y = torch.randn(32, 16, generator=g) * 0.5 + 0.1
+    X8 = ops.quantize_fp8(X)
+    Y8 = ops.quantize_fp8(Y)
+    C = torch.empty(16, 16, dtype=torch.float32, device=gpu)
+    hip.mfma_probe_fp8(X8.to(gpu), Y8.to(gpu), C, _stream())
+    torch.cuda.synchronize()
+    ref = ops.dequantize_fp8(X8).T @ ops.dequantize_fp8(Y8)
+    assert torch.allclose(C.cpu(), ref, atol=1e-4, rtol=1e-4), \
+        (C.cpu() - ref).abs().max()

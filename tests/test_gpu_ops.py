@@ -167,11 +167,15 @@ def test_ldl_wave_solver_parity(gpu, k):
     b = torch.randn(B, k, generator=g).to(gpu)
     x = torch.empty_like(b)
     xb = torch.empty(B, k, dtype=torch.bfloat16, device=gpu)
-    hip.ldl_solve_wave(A, b, x, xb, torch.cuda.current_stream().cuda_stream)
+    x8 = torch.empty(B, k, dtype=torch.uint8, device=gpu)
+    st = torch.cuda.current_stream().cuda_stream
+    hip.ldl_solve_wave(A, b, x, xb, x8, st)
     x_ref = R.cholesky_solve_reference(A.cpu(), b.cpu()).to(gpu)
     torch.cuda.synchronize()
     assert torch.allclose(x, x_ref, atol=1e-3, rtol=1e-3)
     assert torch.allclose(xb.to(torch.float32), x, atol=1e-1, rtol=2e-2)
+    # the e4m3 image must be the RNE quantization of the fp32 solution
+    assert torch.equal(x8.cpu(), ops.quantize_fp8(x.cpu()))
 
 
 def test_slabbed_solve_matches_unslabbed(gpu):
@@ -242,3 +246,107 @@ def test_solve_rank10_padding(gpu):
     torch.cuda.synchronize()
     err = (out.cpu() - ref).abs().amax() / ref.abs().amax()
     assert err < 2e-2, f"rank-10 rel err {err}"
+
+
+# ------------------------------------------------------------ fp8 e4m3 path
+
+@pytest.mark.parametrize("k", [16, 32, 64, 128])
+def test_gramian_fp8_parity(gpu, k):
+    """fp8 gather Gramian vs fp32 reference on the SAME e4m3-quantized
+    factors.  A is rating-value-independent; b uses the e4m3 hi/lo rating
+    pair, so its reference uses the pair-dequantized values."""
+    csr = _rand_csr(rows=300, cols=200, nnz=20_000, seed=k + 7, device=gpu)
+    fac = (torch.randn(200, k, generator=torch.Generator().manual_seed(6))
+           * 0.5)
+    f8 = ops.quantize_fp8(fac)
+    A, b = ops.gramian(csr, f8.to(gpu), reg=0.7)
+    cpu_csr = csr.to("cpu")
+    A_ref, _ = R.gramian_reference(cpu_csr, ops.dequantize_fp8(f8), reg=0.7)
+    pair_csr = CSR(cpu_csr.indptr, cpu_csr.indices,
+                   ops.fp8_rating_pair(cpu_csr.values),
+                   cpu_csr.num_rows, cpu_csr.num_cols)
+    _, b_ref = R.gramian_reference(pair_csr, ops.dequantize_fp8(f8), reg=0.7)
+    torch.cuda.synchronize()
+    scale = A_ref.abs().amax()
+    assert (A.cpu() - A_ref).abs().amax() / scale < 3e-3, \
+        f"A mismatch {(A.cpu() - A_ref).abs().amax()} vs scale {scale}"
+    bscale = b_ref.abs().amax()
+    assert (b.cpu() - b_ref).abs().amax() / bscale < 3e-3
+
+
+@pytest.mark.parametrize("k", [64, 128])
+def test_als_solve_side_fp8_parity(gpu, k):
+    """Full fp8 half-iteration (gramian -> LDL solve -> e4m3 image) vs the
+    fp32 reference solve on dequantized factors + pair ratings."""
+    csr = _rand_csr(rows=400, cols=250, nnz=30_000, seed=k + 11, device=gpu)
+    fac = (torch.randn(250, k, generator=torch.Generator().manual_seed(8))
+           * 0.5)
+    f8 = ops.quantize_fp8(fac)
+    out8 = torch.empty(csr.num_rows, k, dtype=torch.uint8, device=gpu)
+    out = ops.als_solve_side(csr, f8.to(gpu), reg=0.4, out_fp8=out8)
+    cpu_csr = csr.to("cpu")
+    pair_csr = CSR(cpu_csr.indptr, cpu_csr.indices,
+                   ops.fp8_rating_pair(cpu_csr.values),
+                   cpu_csr.num_rows, cpu_csr.num_cols)
+    ref = R.als_solve_side_reference(pair_csr, ops.dequantize_fp8(f8), reg=0.4)
+    torch.cuda.synchronize()
+    err = (out.cpu() - ref).abs().amax()
+    assert err < 5e-3 * max(1.0, float(ref.abs().amax())), err
+    # the emitted e4m3 image quantizes the fp32 solution (RNE)
+    assert torch.equal(out8.cpu(), ops.quantize_fp8(out.cpu()))
+
+
+def test_fused_fp8_matches_modular_fp8(gpu):
+    csr = _rand_csr(rows=500, cols=300, nnz=40_000, seed=91, device=gpu)
+    fac = (torch.randn(300, 64, generator=torch.Generator().manual_seed(12))
+           * 0.5)
+    f8 = ops.quantize_fp8(fac).to(gpu)
+    out_mod = ops.als_solve_side(csr, f8, reg=0.9)
+    out_fused = ops.als_solve_side(csr, f8, reg=0.9, fused=True)
+    torch.cuda.synchronize()
+    assert torch.allclose(out_mod, out_fused, atol=2e-3, rtol=2e-3), \
+        (out_mod - out_fused).abs().max()
+
+
+def test_als_gpu_fp8_end_to_end(gpu):
+    """fp8 factor exchange end to end: training quality within a few
+    percent of the bf16 run (the fp8 convergence-study contract)."""
+    from flink_ms_amd.data.ratings import RatingsShape, synthetic_ratings
+    from flink_ms_amd.models.als import ALSConfig, ALSTrainer
+    from flink_ms_amd.models.mse import evaluate_mse
+    u, i, r = synthetic_ratings(RatingsShape(3000, 1000, 120_000), seed=13)
+    res = {}
+    for fd in ("bf16", "fp8"):
+        tr = ALSTrainer(ALSConfig(iterations=5, num_factors=16,
+                                  lambda_=0.02, factor_dtype=fd))
+        tr.ctx.device = gpu
+        tr.setup(u.long(), i.long(), r, 3000, 1000)
+        tr.fit()
+        m = tr.model()
+        res[fd] = evaluate_mse(m.user_factors.to(gpu),
+                               m.item_factors.to(gpu), u, i, r).mse
+    assert res["fp8"] < res["bf16"] * 1.10 + 1e-3, res
+
+
+def test_gramian_row_order_invariance(gpu):
+    """The degree-descending schedule only permutes the LAUNCH order: the
+    scattered A/b must equal the unordered run exactly."""
+    csr = _rand_csr(rows=350, cols=200, nnz=25_000, seed=55, device=gpu)
+    fac = (torch.randn(200, 64, generator=torch.Generator().manual_seed(9))
+           * 0.5).to(torch.bfloat16).to(gpu)
+    import flink_ms_amd._hip_ops as hip
+    st = torch.cuda.current_stream().cuda_stream
+    k = 64
+    fp = fac.contiguous()
+    A0 = torch.empty(csr.num_rows, k, k, dtype=torch.float32, device=gpu)
+    b0 = torch.empty(csr.num_rows, k, dtype=torch.float32, device=gpu)
+    empty = torch.empty(0, device=gpu)
+    hip.gramian(csr.indptr, csr.indices, csr.values, fp, A0, b0, empty,
+                0.5, st)
+    order = torch.argsort(csr.row_counts(), descending=True).to(torch.int32)
+    A1 = torch.empty_like(A0)
+    b1 = torch.empty_like(b0)
+    hip.gramian(csr.indptr, csr.indices, csr.values, fp, A1, b1,
+                order.to(gpu), 0.5, st)
+    torch.cuda.synchronize()
+    assert torch.equal(A0, A1) and torch.equal(b0, b1)

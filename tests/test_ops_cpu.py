@@ -166,3 +166,49 @@ def test_libsvm_roundtrip(tmp_path):
     assert torch.equal(csr2.indptr, csr.indptr)
     assert torch.equal(csr2.indices, csr.indices)   # back to 0-based
     assert torch.allclose(csr2.values, csr.values, atol=1e-6)
+
+
+# ------------------------------------------------------------ fp8 e4m3
+
+def test_fp8_quant_roundtrip():
+    from flink_ms_amd import ops
+    t = torch.tensor([0.0, 1.0, -1.0, 0.3, 448.0, 1e-4, -0.0517])
+    q = ops.quantize_fp8(t)
+    assert q.dtype == torch.uint8
+    d = ops.dequantize_fp8(q)
+    # idempotent: every dequantized value is exactly representable
+    assert torch.equal(ops.quantize_fp8(d), q)
+    # e4m3 relative step is 2^-4 at worst for normal values
+    n = t.abs() > 2 ** -6
+    assert ((d[n] - t[n]).abs() / t[n].abs()).max() < 2 ** -4
+
+
+def test_fp8_rating_pair_accuracy():
+    from flink_ms_amd import ops
+    g = torch.Generator().manual_seed(3)
+    r = torch.rand(10_000, generator=g) * 4.5 + 0.5  # rating-like range
+    eff = ops.fp8_rating_pair(r)
+    rel = ((eff - r).abs() / r).max()
+    # the r1 study's measured bound: max relative error ~1.9e-3
+    assert rel < 4e-3, rel
+
+
+def test_als_trainer_fp8_cpu_emulation():
+    """factor_dtype='fp8' on CPU runs the dequantize->reference->requantize
+    emulation: same quantization semantics as the GPU kernels, so training
+    quality must stay within a few percent of the fp32 run."""
+    from flink_ms_amd.data.ratings import RatingsShape, synthetic_ratings
+    from flink_ms_amd.models.als import ALSConfig, ALSTrainer
+    from flink_ms_amd.models.mse import evaluate_mse
+    u, i, r = synthetic_ratings(RatingsShape(300, 120, 6000), seed=4)
+    res = {}
+    for fd in ("bf16", "fp8"):
+        tr = ALSTrainer(ALSConfig(iterations=4, num_factors=8, lambda_=0.05,
+                                  dtype=torch.float32, factor_dtype=fd))
+        tr.setup(u.long(), i.long(), r, 300, 120)
+        tr.fit()
+        m = tr.model()
+        res[fd] = evaluate_mse(m.user_factors, m.item_factors, u, i, r).mse
+        if fd == "fp8":
+            assert tr.user_shard.dtype == torch.uint8
+    assert res["fp8"] < res["bf16"] * 1.15 + 1e-3, res
