@@ -119,14 +119,14 @@ __global__ void k_als_solve_fused_fp8(const long long* __restrict__ indptr,
                                       const int* __restrict__ row_order,
                                       long long nrows, float reg) {
     constexpr int K = Geo<KT>::K;
-    __shared__ __align__(16) char smem[Geo<KT>::SMEM_TRI];
+    __shared__ __align__(16) char smem[Geo<KT>::SMEM8_TRI];
     long long row = blockIdx.x;
     if (row >= nrows) return;
     if (row_order) row = row_order[row];
 
-    const int n = gramian_to_lds_fp8<KT, true, true>(smem, indptr, indices,
-                                                     values, factors, row,
-                                                     reg);
+    const int n = gramian_to_lds<KT, true, true, true>(smem, indptr,
+                                                       indices, values,
+                                                       factors, row, reg);
     if (n == 0) {
         for (int c = threadIdx.x; c < K; c += 256) {
             out_f32[row * K + c] = 0.0f;
@@ -152,15 +152,14 @@ __global__ void k_gramian_fp8(const long long* __restrict__ indptr,
                               const int* __restrict__ row_order,
                               long long nrows, float reg) {
     constexpr int K = Geo<KT>::K;
-    static_assert(Geo<KT>::STAGE8_BYTES <= Geo<KT>::SMEM, "stage fits");
-    __shared__ __align__(16) char smem[Geo<KT>::SMEM];
+    __shared__ __align__(16) char smem[Geo<KT>::SMEM8];
     long long row = blockIdx.x;
     if (row >= nrows) return;
     if (row_order) row = row_order[row];
     const int tid = threadIdx.x;
 
-    const int n = gramian_to_lds_fp8<KT>(smem, indptr, indices, values,
-                                         factors, row, reg);
+    const int n = gramian_to_lds<KT, true, false, true>(
+        smem, indptr, indices, values, factors, row, reg);
     float* A = (float*)smem;
     float* b = A + K * (K + 1);
     if (n == 0) {

@@ -20,8 +20,10 @@ __global__ void k_stage_dump(const long long* __restrict__ indptr,
     __shared__ __align__(16) char smem[Geo<KT>::SMEM];
     const long long p0 = indptr[0];
     const int n = (int)(indptr[1] - p0);
-    stage_zero_pad<KT>(smem);
-    stage_chunk<KT>(smem, indices, values, factors, p0, n);
+    stage_zero_pad_all<KT, false, 1>(smem);
+    if (threadIdx.x < 64)
+        stage_chunk_w<KT, false>(smem, indices, values, factors, p0, n,
+                                 threadIdx.x);
     __syncthreads();
     for (int i = threadIdx.x; i < (K + 16) * 32; i += 256) {
         const int row = i / 32, col = i % 32;
@@ -41,12 +43,14 @@ __global__ void k_frag_dump(const long long* __restrict__ indptr,
     __shared__ __align__(16) char smem[Geo<KT>::SMEM];
     const long long p0 = indptr[0];
     const int n = (int)(indptr[1] - p0);
-    stage_zero_pad<KT>(smem);
-    stage_chunk<KT>(smem, indices, values, factors, p0, n);
+    stage_zero_pad_all<KT, false, 1>(smem);
+    if (threadIdx.x < 64)
+        stage_chunk_w<KT, false>(smem, indices, values, factors, p0, n,
+                                 threadIdx.x);
     __syncthreads();
     const int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
     bf16x8 frag[KT + 1];
-    read_frags<KT>(smem, lane, frag);
+    read_frags<KT, false>(smem, lane, frag);
     for (int t = 0; t <= KT; ++t)
         for (int j = 0; j < 8; ++j)
             out[((w * (KT + 1) + t) * 64 + lane) * 8 + j] =
