@@ -136,6 +136,16 @@ def als_solve_side(
             else:
                 ops.als_solve_fused(csr.indptr, csr.indices, csr.values, fac,
                                     out, ob, ro, float(reg), _stream())
+        elif slab_rows is None:
+            # k<=64 default: wave-fused Gramian+LDL — one wave per entity,
+            # A lives only in MFMA C-fragments (no nrows*k*k HBM round
+            # trip, no barriers; the r2 profile showed the modular path
+            # bound by exactly that A write+read traffic)
+            out = torch.empty(csr.num_rows, k, dtype=torch.float32,
+                              device=fac.device)
+            ops.als_solve_wavefused(csr.indptr, csr.indices, csr.values,
+                                    fac, out, ob, o8, ro, float(reg),
+                                    _stream())
         else:
             # slab the normal equations: A is nrows*k*k fp32, which at the
             # 1B-rating configs would exceed HBM if materialized whole.

@@ -745,6 +745,187 @@ extern "C" hipError_t fma_ldl_solve_wave_reg(
     return hipGetLastError();
 }
 
+// -------- wave-fused ALS solve (k <= 64): Gramian + LDL in ONE wave --------
+// r2 profiling: with the v2 multi-chunk gramian the modular path became
+// A-round-trip bound — the gramian writes nrows*K*K fp32 to HBM (~2.9 TB/s)
+// and the wave solver reads the lower tiles back (~1.5 TB/s).  This kernel
+// removes that traffic entirely: one WAVE owns one entity end to end and
+// the Gramian MFMAs accumulate DIRECTLY into the lower-triangle C-fragment
+// tiles T[] that the register LDL (wreg_panels/forward/backward) consumes.
+// No barriers anywhere (wave-local LDS + WREG_FENCE ordering); gather
+// stalls of one wave overlap solver compute of the others.
+//
+// Tile algebra: lower tile (I,J) = mfma(frag[I], frag[J]) — D[row][col] =
+// sum_r G[r][I*16+row]*G[r][J*16+col] = A[I*16+row][J*16+col], exactly the
+// (g4*4+r, li) layout wreg_load_A produced in the modular path.  b comes
+// from KT EXT tiles (cols 0/1 = rating hi/lo), redistributed to
+// x0 = b[lane] through a 128-float LDS bounce.
+
+template <int KT, bool FP8, int t = 0>
+DEV_INLINE void mfma_lower_tiles(const typename FragT<FP8>::type* frag,
+                                 f32x4* T) {
+    if constexpr (t < KT * (KT + 1) / 2) {
+        constexpr int I = lo_tile_i(t);
+        constexpr int J = t - (I * (I + 1)) / 2;
+        if constexpr (FP8)
+            T[t] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+                frag[I], frag[J], T[t], 0, 0, 0);
+        else
+            T[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                frag[I], frag[J], T[t], 0, 0, 0);
+        mfma_lower_tiles<KT, FP8, t + 1>(frag, T);
+    }
+}
+
+template <int KT, bool FP8, int P = 0>
+DEV_INLINE void mfma_ext_tiles(const typename FragT<FP8>::type* frag,
+                               f32x4* E) {
+    if constexpr (P < KT) {
+        if constexpr (FP8)
+            E[P] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+                frag[P], frag[KT], E[P], 0, 0, 0);
+        else
+            E[P] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                frag[P], frag[KT], E[P], 0, 0, 0);
+        mfma_ext_tiles<KT, FP8, P + 1>(frag, E);
+    }
+}
+
+template <int KT, int P = 0>
+DEV_INLINE void wavefused_dump_b(const f32x4* E, float* scr, int g4, int li) {
+    if constexpr (P < KT) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+            if (li == 0) scr[P * 16 + g4 * 4 + r] = E[P][r];
+            else if (li == 1) scr[64 + P * 16 + g4 * 4 + r] = E[P][r];
+        }
+        wavefused_dump_b<KT, P + 1>(E, scr, g4, li);
+    }
+}
+
+template <int KT, int I = 0>
+DEV_INLINE void wavefused_diag(f32x4* T, float regn, int g4, int li) {
+    if constexpr (I < KT) {
+        constexpr int t = tri_off(I, I);
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+            if (li == g4 * 4 + r) {
+                const float dd = T[t][r] + regn;
+                T[t][r] = dd <= 0.0f ? 1.0f : dd;   // degenerate guard
+            }
+        wavefused_diag<KT, I + 1>(T, regn, g4, li);
+    }
+}
+
+template <int KT, bool FP8>
+__launch_bounds__(256)
+__global__ void k_als_solve_wavefused(const long long* __restrict__ indptr,
+                                      const int* __restrict__ indices,
+                                      const float* __restrict__ values,
+                                      const void* __restrict__ factors,
+                                      float* __restrict__ out_f32,
+                                      unsigned short* __restrict__ out_bf16,
+                                      unsigned char* __restrict__ out_fp8,
+                                      const int* __restrict__ row_order,
+                                      long long nrows, float reg) {
+    constexpr int K = KT * 16;
+    static_assert(K <= 64, "wave-fused path handles k <= 64");
+    constexpr int NA = KT * (KT + 1) / 2;
+    constexpr int TROW = FP8 ? Geo<KT>::TROW8 : Geo<KT>::TROW;
+    constexpr int SB = (K + 16) * TROW;            // stage bytes per wave
+    constexpr int SCRF = (K * 17 > 1040) ? K * 17 : 1040;  // solver scratch
+    constexpr int WB = (SB > SCRF * 4) ? SB : SCRF * 4;    // union, per wave
+    __shared__ __align__(16) char smem[4 * WB];
+    const int w = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
+    const int lane = threadIdx.x & 63;
+    const long long e0 = (long long)blockIdx.x * 4 + w;
+    if (e0 >= nrows) return;
+    const long long e = row_order ? row_order[e0] : e0;
+    char* buf = smem + (long long)w * WB;
+    float* scr = (float*)buf;          // union: stage dies before the solve
+    const int g4 = lane >> 4, li = lane & 15;
+    const long long p0 = indptr[e];
+    const int n = (int)(indptr[e + 1] - p0);
+    if (n == 0) {
+        if (lane < K) {
+            out_f32[e * K + lane] = 0.0f;
+            if (out_bf16) out_bf16[e * K + lane] = 0;
+            if (out_fp8) out_fp8[e * K + lane] = 0;
+        }
+        return;
+    }
+    {   // zero my buffer's EXT pad rows K+2..K+15 (wave-local)
+        constexpr int SEGS = FP8 ? 8 : 16;
+        for (int i = lane; i < 14 * SEGS; i += WAVE) {
+            const int row = K + 2 + i / SEGS, seg = i % SEGS;
+            *(unsigned*)(buf + (long long)row * TROW + seg * 4) = 0u;
+        }
+    }
+    f32x4 T[NA], E[KT];
+#pragma unroll
+    for (int t = 0; t < NA; ++t) T[t] = f32x4{0, 0, 0, 0};
+#pragma unroll
+    for (int p = 0; p < KT; ++p) E[p] = f32x4{0, 0, 0, 0};
+    const int nchunks = (n + 31) >> 5;
+    for (int ch = 0; ch < nchunks; ++ch) {
+        stage_chunk_w<KT, FP8>(buf, indices, values, factors,
+                               p0 + (long long)ch * 32, n - ch * 32, lane);
+        WREG_FENCE();                  // cross-lane LDS write -> read
+        typename FragT<FP8>::type frag[KT + 1];
+        read_frags<KT, FP8>(buf, lane, frag);
+        WREG_FENCE();                  // reads drained before next stage
+        mfma_lower_tiles<KT, FP8>(frag, T);
+        mfma_ext_tiles<KT, FP8>(frag, E);
+    }
+    // b = bhi + blo redistributed to lane = row; then lambda*n*I
+    wavefused_dump_b<KT>(E, scr, g4, li);
+    WREG_FENCE();
+    float x0 = (lane < K) ? scr[lane] + scr[64 + lane] : 0.0f;
+    WREG_FENCE();                      // b read before panel dumps reuse scr
+    wavefused_diag<KT>(T, reg * (float)n, g4, li);
+    // in-register LDL + substitution (shared with k_ldl_solve_wave_reg)
+    float d0 = 1.0f;
+    wreg_panels<KT, 0>(T, scr, lane, g4, li, d0);
+    const float id0 = d0 > 0.0f ? 1.0f / d0 : 0.0f;
+    wreg_forward<KT, 0>(T, scr, x0, id0, lane, g4, li);
+    x0 *= id0;
+    wreg_backward<KT, KT - 1>(T, scr, x0, id0, lane, g4, li);
+    if (lane < K) {
+        out_f32[e * K + lane] = x0;
+        if (out_bf16) out_bf16[e * K + lane] = f2bf(x0);
+        if (out_fp8) out_fp8[e * K + lane] = f2fp8(x0);
+    }
+}
+
+extern "C" hipError_t fma_als_solve_wavefused(
+    int k, int fp8, const long long* indptr, const int* indices,
+    const float* values, const void* factors, float* out_f32,
+    unsigned short* out_bf16, unsigned char* out_fp8, const int* row_order,
+    long long nrows, float reg, hipStream_t stream) {
+    if (k % 16 || k < 16 || k > 64 || nrows <= 0) return hipErrorInvalidValue;
+    dim3 grid((unsigned)((nrows + 3) / 4)), block(256);
+#define WF_CASE(KT)                                                           \
+    case KT:                                                                  \
+        if (fp8)                                                              \
+            k_als_solve_wavefused<KT, true><<<grid, block, 0, stream>>>(      \
+                indptr, indices, values, factors, out_f32, out_bf16,          \
+                out_fp8, row_order, nrows, reg);                              \
+        else                                                                  \
+            k_als_solve_wavefused<KT, false><<<grid, block, 0, stream>>>(     \
+                indptr, indices, values, factors, out_f32, out_bf16,          \
+                out_fp8, row_order, nrows, reg);                              \
+        break;
+    switch (k / 16) {
+        WF_CASE(1)
+        WF_CASE(2)
+        WF_CASE(3)
+        WF_CASE(4)
+        default: return hipErrorInvalidValue;
+    }
+#undef WF_CASE
+    return hipGetLastError();
+}
+
 extern "C" hipError_t fma_ldl_solve_wave(
     int k, const float* A_in, const float* b_in, float* x_out,
     unsigned short* x_bf16, unsigned char* x_fp8, long long nrows,

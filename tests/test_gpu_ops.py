@@ -350,3 +350,30 @@ def test_gramian_row_order_invariance(gpu):
                 order.to(gpu), 0.5, st)
     torch.cuda.synchronize()
     assert torch.equal(A0, A1) and torch.equal(b0, b1)
+
+
+@pytest.mark.parametrize("dtype", ["bf16", "fp8"])
+def test_wavefused_matches_modular(gpu, dtype):
+    """The wave-fused Gramian+LDL (A in registers, no HBM round trip) must
+    reproduce the modular gramian->ldl_solve_wave_reg path."""
+    csr = _rand_csr(rows=403, cols=250, nnz=35_000, seed=71, device=gpu)
+    fac32 = (torch.randn(250, 64, generator=torch.Generator().manual_seed(3))
+             * 0.5)
+    fac = (ops.quantize_fp8(fac32) if dtype == "fp8"
+           else fac32.to(torch.bfloat16)).to(gpu)
+    out_wf = ops.als_solve_side(csr, fac, reg=0.3)            # wavefused
+    out_mod = ops.als_solve_side(csr, fac, reg=0.3,
+                                 slab_rows=csr.num_rows)      # modular
+    torch.cuda.synchronize()
+    assert torch.allclose(out_wf, out_mod, atol=1e-4, rtol=1e-4), \
+        (out_wf - out_mod).abs().max()
+    # zero-degree rows (row 402 likely has ratings; force an empty CSR row)
+    import flink_ms_amd._hip_ops as hip
+    e8 = torch.empty(0, device=gpu)
+    indptr = torch.tensor([0, 0, csr.nnz], dtype=torch.int64, device=gpu)
+    out2 = torch.empty(2, 64, dtype=torch.float32, device=gpu)
+    hip.als_solve_wavefused(indptr, csr.indices, csr.values, fac, out2,
+                            e8, e8, e8, 0.3,
+                            torch.cuda.current_stream().cuda_stream)
+    torch.cuda.synchronize()
+    assert out2[0].abs().sum() == 0.0
