@@ -54,6 +54,11 @@ int fma_als_solve_wavefused(int k, int fp8, const int64_t* indptr,
                             unsigned short* out_bf16, unsigned char* out_fp8,
                             const int* row_order, int64_t nrows, float reg,
                             void* stream);
+int fma_als_solve_wavefused2(int k, const int64_t* indptr,
+                             const int* indices, const float* values,
+                             const unsigned char* factors, float* out_f32,
+                             unsigned char* out_fp8, const int* row_order,
+                             int64_t nrows, float reg, void* stream);
 int fma_ldl_solve_wave(int k, const float* A_in, const float* b_in,
                        float* x_out, unsigned short* x_bf16,
                        unsigned char* x_fp8, int64_t nrows, void* stream);
@@ -271,6 +276,34 @@ void als_solve_wavefused(torch::Tensor indptr, torch::Tensor indices,
               "als_solve_wavefused");
 }
 
+// wave-pair fused path for 64 < k <= 128 (fp8 gathers only)
+void als_solve_wavefused2(torch::Tensor indptr, torch::Tensor indices,
+                          torch::Tensor values, torch::Tensor factors,
+                          torch::Tensor out_f32, torch::Tensor out_fp8,
+                          torch::Tensor row_order, double reg,
+                          int64_t stream) {
+    check_t(indptr, torch::kInt64, "indptr");
+    check_t(indices, torch::kInt32, "indices");
+    check_t(values, torch::kFloat32, "values");
+    check_t(factors, torch::kUInt8, "factors");
+    check_t(out_f32, torch::kFloat32, "out_f32");
+    const int k = (int)factors.size(1);
+    const long long nrows = indptr.size(0) - 1;
+    TORCH_CHECK(out_f32.size(0) == nrows && out_f32.size(1) == k,
+                "out_f32 shape mismatch");
+    unsigned char* o8 = nullptr;
+    if (out_fp8.numel() > 0) {
+        TORCH_CHECK(out_fp8.numel() == out_f32.numel(), "out_fp8 shape");
+        o8 = fp8_ptr_mut(out_fp8);
+    }
+    check_hip(fma_als_solve_wavefused2(
+                  k, indptr.data_ptr<int64_t>(), indices.data_ptr<int>(),
+                  values.data_ptr<float>(), fp8_ptr(factors),
+                  out_f32.data_ptr<float>(), o8, order_ptr(row_order, nrows),
+                  nrows, (float)reg, (void*)stream),
+              "als_solve_wavefused2");
+}
+
 void ldl_solve_wave(torch::Tensor A, torch::Tensor b, torch::Tensor x,
                     torch::Tensor x_bf16, torch::Tensor x_fp8,
                     int64_t stream) {
@@ -458,6 +491,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("gramian_fp8", &gramian_fp8);
     m.def("als_solve_fused_fp8", &als_solve_fused_fp8);
     m.def("als_solve_wavefused", &als_solve_wavefused);
+    m.def("als_solve_wavefused2", &als_solve_wavefused2);
     m.def("cholesky_solve", &cholesky_solve);
     m.def("cholesky_solve_ph", &cholesky_solve_ph);
     m.def("ldl_solve_wave", &ldl_solve_wave);

@@ -385,3 +385,31 @@ def test_wavefused_matches_modular(gpu, dtype, k):
                             torch.cuda.current_stream().cuda_stream)
     torch.cuda.synchronize()
     assert out2[0].abs().sum() == 0.0
+
+
+@pytest.mark.parametrize("k", [80, 128])
+def test_wavefused2_matches_block_fused(gpu, k):
+    """The wave-pair register LDL (64<k<=128) must reproduce the
+    block-fused kernel, including odd tails and empty rows."""
+    csr = _rand_csr(rows=401, cols=250, nnz=40_000, seed=k + 3, device=gpu)
+    fac = ops.quantize_fp8(
+        torch.randn(250, k, generator=torch.Generator().manual_seed(2))
+        * 0.5).to(gpu)
+    out8 = torch.empty(csr.num_rows, k, dtype=torch.uint8, device=gpu)
+    out_wp = ops.als_solve_side(csr, fac, reg=0.6, out_fp8=out8)
+    out_bl = ops.als_solve_side(csr, fac, reg=0.6, fused=True)
+    torch.cuda.synchronize()
+    assert torch.allclose(out_wp, out_bl, atol=1e-4, rtol=1e-4), \
+        (out_wp - out_bl).abs().max()
+    assert torch.equal(out8.cpu(), ops.quantize_fp8(out_wp.cpu()))
+    # empty row + odd nrows tail
+    indptr = torch.tensor([0, 0, csr.nnz, csr.nnz], dtype=torch.int64,
+                          device=gpu)
+    out3 = torch.empty(3, k, dtype=torch.float32, device=gpu)
+    import flink_ms_amd._hip_ops as hip
+    e = torch.empty(0, device=gpu)
+    hip.als_solve_wavefused2(indptr, csr.indices, csr.values, fac, out3,
+                             e, e, 0.6, torch.cuda.current_stream().cuda_stream)
+    torch.cuda.synchronize()
+    assert out3[0].abs().sum() == 0 and out3[2].abs().sum() == 0
+    assert torch.isfinite(out3[1]).all()
