@@ -387,10 +387,12 @@ def test_wavefused_matches_modular(gpu, dtype, k):
     assert out2[0].abs().sum() == 0.0
 
 
-@pytest.mark.parametrize("k", [80, 128])
+@pytest.mark.parametrize("k", [80, 96, 128])
 def test_wavefused2_matches_block_fused(gpu, k):
-    """The wave-pair register LDL (64<k<=128) must reproduce the
-    block-fused kernel, including odd tails and empty rows."""
+    """The wave-pair register LDL (64<k<=128) and the block-fused kernel
+    must BOTH match the fp32 reference on the same quantized inputs
+    (comparing against ground truth pinpoints which kernel is wrong when
+    they disagree), plus odd tails and empty rows."""
     csr = _rand_csr(rows=401, cols=250, nnz=40_000, seed=k + 3, device=gpu)
     fac = ops.quantize_fp8(
         torch.randn(250, k, generator=torch.Generator().manual_seed(2))
@@ -398,9 +400,18 @@ def test_wavefused2_matches_block_fused(gpu, k):
     out8 = torch.empty(csr.num_rows, k, dtype=torch.uint8, device=gpu)
     out_wp = ops.als_solve_side(csr, fac, reg=0.6, out_fp8=out8)
     out_bl = ops.als_solve_side(csr, fac, reg=0.6, fused=True)
+    cpu_csr = csr.to("cpu")
+    pair_csr = CSR(cpu_csr.indptr, cpu_csr.indices,
+                   ops.fp8_rating_pair(cpu_csr.values),
+                   cpu_csr.num_rows, cpu_csr.num_cols)
+    ref = R.als_solve_side_reference(pair_csr,
+                                     ops.dequantize_fp8(fac.cpu()), reg=0.6)
     torch.cuda.synchronize()
-    assert torch.allclose(out_wp, out_bl, atol=1e-4, rtol=1e-4), \
-        (out_wp - out_bl).abs().max()
+    scale = max(1.0, float(ref.abs().amax()))
+    err_wp = (out_wp.cpu() - ref).abs().amax() / scale
+    err_bl = (out_bl.cpu() - ref).abs().amax() / scale
+    assert err_wp < 5e-3, f"wave-pair vs reference: {err_wp}"
+    assert err_bl < 5e-3, f"block-fused vs reference: {err_bl}"
     assert torch.equal(out8.cpu(), ops.quantize_fp8(out_wp.cpu()))
     # empty row + odd nrows tail
     indptr = torch.tensor([0, 0, csr.nnz, csr.nnz], dtype=torch.int64,
