@@ -1142,6 +1142,10 @@ DEV_INLINE void wp_panels(f32x4* T, float* scr, int lane, int g4, int li,
             if ((lane >> 4) + 4 == PI)
                 db = scr[(lane + 64 - P0) * 17 + (lane + 64 - P0)];
         }
+        // load the FACTORED panel column back into the register tiles (the
+        // substitutions dump L from T; without this they would read raw A)
+        if (p == 0) wp_load_col<KT, PI, 0>(T, scr, g4, li);
+        else        wp_load_col<KT, PI, 1>(T, scr, g4, li);
         if constexpr (PI + 1 < KT) {
             float ndk[4];
 #pragma unroll
