@@ -966,112 +966,110 @@ extern "C" const char* fma_err_str(int err) {
     return hipGetErrorString((hipError_t)err);
 }
 
-// ------- wave-PAIR fused ALS solve (64 < k <= 128, fp8 gathers) -------
+// ------- wave-QUAD fused ALS solve (64 < k <= 128, fp8 gathers) -------
 // The k=128 block-fused kernel serializes its LDL at Gramian occupancy and
 // the modular path pays an 80 GB A round trip; this kernel extends the
-// wave-fused design: TWO waves own one entity, each holding half the
-// lower-triangle MFMA C-fragment tiles (parity split: wave p owns tiles
-// t % 2 == p).  The Gramian accumulates straight into those registers
-// (2 staged chunks in flight per entity pair), then an in-register LDL
-// runs with block barriers only at phase edges: panel columns bounce
-// through a per-pair LDS scratch, 16-col panels factor on wave 0 of the
-// pair (2 rows/lane), trailing updates run on BOTH waves' own tiles via
-// f32 MFMA, and the substitutions keep x as TWO registers per lane
-// (rows lane and lane+64) redundantly on each wave so the serial chain
-// needs no cross-wave traffic.  Barrier counts are compile-time uniform
-// across the block; the gather-round count is MAX-reduced over the two
-// entities (pair with fewer chunks idles — pair entities are
-// degree-adjacent under row_order, so the waste is small).
+// k<=64 wave-fused design: all FOUR waves of a block own ONE entity, each
+// holding a quarter of the lower-triangle MFMA C-fragment tiles (t % 4 ==
+// wave).  The Gramian accumulates straight into those registers (4 staged
+// chunks in flight per entity), then an in-register LDL runs with block
+// barriers at phase edges: panel columns bounce through LDS scratch,
+// 16-col panels factor on wave 0 (2 rows/lane), trailing updates run on
+// every wave's own tiles via f32 MFMA, and the substitutions keep x as
+// TWO registers per lane (rows lane and lane+64) redundantly on each wave
+// so the serial chain needs no cross-wave traffic.  One entity per block
+// keeps every barrier trivially uniform (n==0 exits whole-block).
 
 template <int KT, int P, int t = 0>
-DEV_INLINE void wp_mfma(const fp8x8* frag, f32x4* T, f32x4* E) {
+DEV_INLINE void wq_mfma(const fp8x8* frag, f32x4* T, f32x4* E) {
     constexpr int NA = KT * (KT + 1) / 2;
     if constexpr (t < NA) {
-        if constexpr ((t & 1) == P) {
+        if constexpr ((t & 3) == P) {
             constexpr int I = lo_tile_i(t);
             constexpr int J = t - (I * (I + 1)) / 2;
-            T[t >> 1] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
-                frag[I], frag[J], T[t >> 1], 0, 0, 0);
+            T[t >> 2] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+                frag[I], frag[J], T[t >> 2], 0, 0, 0);
         }
-        wp_mfma<KT, P, t + 1>(frag, T, E);
+        wq_mfma<KT, P, t + 1>(frag, T, E);
     } else if constexpr (t < NA + KT) {
         constexpr int Pt = t - NA;
-        if constexpr ((Pt & 1) == P)
-            E[Pt >> 1] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
-                frag[Pt], frag[KT], E[Pt >> 1], 0, 0, 0);
-        wp_mfma<KT, P, t + 1>(frag, T, E);
+        if constexpr ((Pt & 3) == P)
+            E[Pt >> 2] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+                frag[Pt], frag[KT], E[Pt >> 2], 0, 0, 0);
+        wq_mfma<KT, P, t + 1>(frag, T, E);
     }
 }
 
 template <int KT, int P, int Pt = 0>
-DEV_INLINE void wp_dump_b(const f32x4* E, float* scr, int g4, int li) {
+DEV_INLINE void wq_dump_b(const f32x4* E, float* scr, int g4, int li) {
     constexpr int K = KT * 16;
     if constexpr (Pt < KT) {
-        if constexpr ((Pt & 1) == P) {
+        if constexpr ((Pt & 3) == P) {
 #pragma unroll
             for (int r = 0; r < 4; ++r) {
-                if (li == 0) scr[Pt * 16 + g4 * 4 + r] = E[Pt >> 1][r];
-                else if (li == 1) scr[K + Pt * 16 + g4 * 4 + r] = E[Pt >> 1][r];
+                if (li == 0) scr[Pt * 16 + g4 * 4 + r] = E[Pt >> 2][r];
+                else if (li == 1)
+                    scr[K + Pt * 16 + g4 * 4 + r] = E[Pt >> 2][r];
             }
         }
-        wp_dump_b<KT, P, Pt + 1>(E, scr, g4, li);
+        wq_dump_b<KT, P, Pt + 1>(E, scr, g4, li);
     }
 }
 
 template <int KT, int P, int I = 0>
-DEV_INLINE void wp_diag(f32x4* T, float regn, int g4, int li) {
+DEV_INLINE void wq_diag(f32x4* T, float regn, int g4, int li) {
     if constexpr (I < KT) {
         constexpr int t = tri_off(I, I);
-        if constexpr ((t & 1) == P) {
+        if constexpr ((t & 3) == P) {
 #pragma unroll
             for (int r = 0; r < 4; ++r)
                 if (li == g4 * 4 + r) {
-                    const float dd = T[t >> 1][r] + regn;
-                    T[t >> 1][r] = dd <= 0.0f ? 1.0f : dd;
+                    const float dd = T[t >> 2][r] + regn;
+                    T[t >> 2][r] = dd <= 0.0f ? 1.0f : dd;
                 }
         }
-        wp_diag<KT, P, I + 1>(T, regn, g4, li);
+        wq_diag<KT, P, I + 1>(T, regn, g4, li);
     }
 }
 
-// dump the PI-th 16-column panel (tiles (I,PI), I>=PI) into scr rows
-// relative to P0, stride 17 — each wave dumps its owned tiles
+// dump / load the PI-th 16-column panel (tiles (I,PI), I>=PI) to/from scr
+// rows relative to P0, stride 17 — each wave handles its owned tiles
 template <int KT, int PI, int P, int I = PI>
-DEV_INLINE void wp_dump_col(const f32x4* T, float* scr, int g4, int li) {
+DEV_INLINE void wq_dump_col(const f32x4* T, float* scr, int g4, int li) {
     if constexpr (I < KT) {
         constexpr int t = tri_off(I, PI);
-        if constexpr ((t & 1) == P) {
+        if constexpr ((t & 3) == P) {
 #pragma unroll
             for (int r = 0; r < 4; ++r)
-                scr[((I - PI) * 16 + g4 * 4 + r) * 17 + li] = T[t >> 1][r];
+                scr[((I - PI) * 16 + g4 * 4 + r) * 17 + li] = T[t >> 2][r];
         }
-        wp_dump_col<KT, PI, P, I + 1>(T, scr, g4, li);
+        wq_dump_col<KT, PI, P, I + 1>(T, scr, g4, li);
     }
 }
 
 template <int KT, int PI, int P, int I = PI>
-DEV_INLINE void wp_load_col(f32x4* T, const float* scr, int g4, int li) {
+DEV_INLINE void wq_load_col(f32x4* T, const float* scr, int g4, int li) {
     if constexpr (I < KT) {
         constexpr int t = tri_off(I, PI);
-        if constexpr ((t & 1) == P) {
+        if constexpr ((t & 3) == P) {
 #pragma unroll
             for (int r = 0; r < 4; ++r)
-                T[t >> 1][r] = scr[((I - PI) * 16 + g4 * 4 + r) * 17 + li];
+                T[t >> 2][r] = scr[((I - PI) * 16 + g4 * 4 + r) * 17 + li];
         }
-        wp_load_col<KT, PI, P, I + 1>(T, scr, g4, li);
+        wq_load_col<KT, PI, P, I + 1>(T, scr, g4, li);
     }
 }
 
 template <int KT, int PI, int P, int RB, int CB>
-DEV_INLINE void wp_trail(f32x4* T, const float* scr, const float* ndk,
+DEV_INLINE void wq_trail(f32x4* T, const float* scr, const float* ndk,
                          int g4, int li) {
     if constexpr (RB < KT) {
         if constexpr (CB > RB) {
-            wp_trail<KT, PI, P, RB + 1, PI + 1>(T, scr, ndk, g4, li);
+            wq_trail<KT, PI, P, RB + 1, PI + 1>(T, scr, ndk, g4, li);
         } else {
             constexpr int t = tri_off(RB, CB);
-            if constexpr ((t & 1) == P) {
-                f32x4 acc = T[t >> 1];
+            if constexpr ((t & 3) == P) {
+                f32x4 acc = T[t >> 2];
 #pragma unroll
                 for (int kk = 0; kk < 4; ++kk) {
                     const int pc = 4 * kk + g4;
@@ -1081,26 +1079,25 @@ DEV_INLINE void wp_trail(f32x4* T, const float* scr, const float* ndk,
                     acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc,
                                                                0, 0, 0);
                 }
-                T[t >> 1] = acc;
+                T[t >> 2] = acc;
             }
-            wp_trail<KT, PI, P, RB, CB + 1>(T, scr, ndk, g4, li);
+            wq_trail<KT, PI, P, RB, CB + 1>(T, scr, ndk, g4, li);
         }
     }
 }
 
-// LDL panel loop: dump column -> factor (wave 0 of the pair, 2 rows/lane)
-// -> trailing MFMA on owned tiles.  Uniform block barriers.
+// LDL panel loop: dump column -> factor (wave 0, 2 rows/lane) -> load the
+// factored column back -> trailing MFMA on owned tiles.  Uniform barriers.
 template <int KT, int P, int PI = 0>
-DEV_INLINE void wp_panels(f32x4* T, float* scr, int lane, int g4, int li,
-                          int p, float& da, float& db) {
+DEV_INLINE void wq_panels(f32x4* T, float* scr, int lane, int g4, int li,
+                          int w, float& da, float& db) {
     if constexpr (PI < KT) {
         constexpr int K = KT * 16;
         constexpr int P0 = 16 * PI;
         constexpr int NR = K - P0;          // rows in this panel image
-        if (p == 0) wp_dump_col<KT, PI, 0>(T, scr, g4, li);
-        else        wp_dump_col<KT, PI, 1>(T, scr, g4, li);
+        wq_dump_col<KT, PI, P>(T, scr, g4, li);
         __syncthreads();
-        if (p == 0) {   // factor the panel slab: lane owns rows rel lane, lane+64
+        if (w == 0) {   // factor: lane owns panel rows rel lane, lane+64
             float sa[16], sb[16];
             const bool av = lane < NR, bv = lane + 64 < NR;
             if (av) {
@@ -1144,8 +1141,7 @@ DEV_INLINE void wp_panels(f32x4* T, float* scr, int lane, int g4, int li,
         }
         // load the FACTORED panel column back into the register tiles (the
         // substitutions dump L from T; without this they would read raw A)
-        if (p == 0) wp_load_col<KT, PI, 0>(T, scr, g4, li);
-        else        wp_load_col<KT, PI, 1>(T, scr, g4, li);
+        wq_load_col<KT, PI, P>(T, scr, g4, li);
         if constexpr (PI + 1 < KT) {
             float ndk[4];
 #pragma unroll
@@ -1153,31 +1149,27 @@ DEV_INLINE void wp_panels(f32x4* T, float* scr, int lane, int g4, int li,
                 const float d = scr[(4 * kk + g4) * 17 + (4 * kk + g4)];
                 ndk[kk] = d > 0.0f ? -1.0f / d : 0.0f;
             }
-            if (p == 0) wp_trail<KT, PI, 0, PI + 1, PI + 1>(T, scr, ndk, g4, li);
-            else        wp_trail<KT, PI, 1, PI + 1, PI + 1>(T, scr, ndk, g4, li);
+            wq_trail<KT, PI, P, PI + 1, PI + 1>(T, scr, ndk, g4, li);
         }
         __syncthreads();   // trailing reads done before next panel's dump
-        wp_panels<KT, P, PI + 1>(T, scr, lane, g4, li, p, da, db);
+        wq_panels<KT, P, PI + 1>(T, scr, lane, g4, li, w, da, db);
     }
 }
 
 // forward substitution: x as 2 regs/lane (rows lane, 64+lane), redundant
-// on both waves; column-J tiles bounce through scr per block of 16
+// on every wave; column-J tiles bounce through scr per block of 16
 template <int KT, int P, int J = 0>
-DEV_INLINE void wp_forward(const f32x4* T, float* scr, float& xa, float& xb,
-                           float ida, float idb, int lane, int g4, int li,
-                           int p) {
+DEV_INLINE void wq_forward(const f32x4* T, float* scr, float& xa, float& xb,
+                           float ida, float idb, int lane, int g4, int li) {
     if constexpr (J < KT) {
         constexpr int K = KT * 16;
         constexpr int B0 = 16 * J;
-        if (p == 0) wp_dump_col<KT, J, 0>(T, scr, g4, li);
-        else        wp_dump_col<KT, J, 1>(T, scr, g4, li);
+        wq_dump_col<KT, J, P>(T, scr, g4, li);
         __syncthreads();
 #pragma unroll
         for (int t = 0; t < 16; ++t) {
-            constexpr int KM1 = K - 1;
             const int j = B0 + t;
-            if (j >= KM1) break;
+            if (j >= K - 1) break;
             float zj;
             if constexpr (B0 < 64) {
                 zj = __shfl(xa, j, WAVE) * __shfl(ida, j, WAVE);
@@ -1189,34 +1181,32 @@ DEV_INLINE void wp_forward(const f32x4* T, float* scr, float& xa, float& xb,
                 xb -= scr[(64 + lane - B0) * 17 + t] * zj;
         }
         __syncthreads();
-        wp_forward<KT, P, J + 1>(T, scr, xa, xb, ida, idb, lane, g4, li, p);
+        wq_forward<KT, P, J + 1>(T, scr, xa, xb, ida, idb, lane, g4, li);
     }
 }
 
 // backward: row image of tile-row I (16 rows x K cols, stride K+1)
 template <int KT, int I, int P, int J = 0>
-DEV_INLINE void wp_dump_row(const f32x4* T, float* scr, int g4, int li) {
+DEV_INLINE void wq_dump_row(const f32x4* T, float* scr, int g4, int li) {
     if constexpr (J <= I) {
         constexpr int K = KT * 16;
         constexpr int t = tri_off(I, J);
-        if constexpr ((t & 1) == P) {
+        if constexpr ((t & 3) == P) {
 #pragma unroll
             for (int r = 0; r < 4; ++r)
-                scr[(g4 * 4 + r) * (K + 1) + J * 16 + li] = T[t >> 1][r];
+                scr[(g4 * 4 + r) * (K + 1) + J * 16 + li] = T[t >> 2][r];
         }
-        wp_dump_row<KT, I, P, J + 1>(T, scr, g4, li);
+        wq_dump_row<KT, I, P, J + 1>(T, scr, g4, li);
     }
 }
 
 template <int KT, int P, int I = KT - 1>
-DEV_INLINE void wp_backward(const f32x4* T, float* scr, float& xa, float& xb,
-                            float ida, float idb, int lane, int g4, int li,
-                            int p) {
+DEV_INLINE void wq_backward(const f32x4* T, float* scr, float& xa, float& xb,
+                            float ida, float idb, int lane, int g4, int li) {
     if constexpr (I >= 0) {
         constexpr int K = KT * 16;
         constexpr int B0 = 16 * I;
-        if (p == 0) wp_dump_row<KT, I, 0>(T, scr, g4, li);
-        else        wp_dump_row<KT, I, 1>(T, scr, g4, li);
+        wq_dump_row<KT, I, P>(T, scr, g4, li);
         __syncthreads();
 #pragma unroll
         for (int t = 15; t >= 0; --t) {
@@ -1229,11 +1219,11 @@ DEV_INLINE void wp_backward(const f32x4* T, float* scr, float& xa, float& xb,
                 xc = __shfl(xa, c, WAVE);
             }
             if (lane < c) xa -= scr[t * (K + 1) + lane] * ida * xc;
-            if (64 + lane < c)
+            if (64 + lane < c && 64 + lane < K)
                 xb -= scr[t * (K + 1) + 64 + lane] * idb * xc;
         }
         __syncthreads();
-        wp_backward<KT, P, I - 1>(T, scr, xa, xb, ida, idb, lane, g4, li, p);
+        wq_backward<KT, P, I - 1>(T, scr, xa, xb, ida, idb, lane, g4, li);
     }
 }
 
@@ -1248,87 +1238,114 @@ __global__ void k_als_solve_wavefused2(const long long* __restrict__ indptr,
                                        const int* __restrict__ row_order,
                                        long long nrows, float reg) {
     constexpr int K = KT * 16;
-    static_assert(K > 64 && K <= 128, "wave-pair path: 64 < k <= 128");
+    static_assert(K > 64 && K <= 128, "wave-quad path: 64 < k <= 128");
     constexpr int NA = KT * (KT + 1) / 2;
-    constexpr int NOWN = (NA + 1) / 2;
-    constexpr int NE = (KT + 1) / 2;
+    constexpr int NOWN = (NA + 3) / 4;
+    constexpr int NE = (KT + 3) / 4;
     constexpr int TROW = Geo<KT>::TROW8;
     constexpr int SB = (K + 16) * TROW;
     constexpr int SCRF = (K * 17 > 16 * (K + 1)) ? K * 17 : 16 * (K + 1);
-    constexpr int PB0 = (2 * SB > SCRF * 4) ? 2 * SB : SCRF * 4;
-    constexpr int PB = (PB0 + 15) & ~15;
-    __shared__ __align__(16) char smem[2 * PB];
-    __shared__ int rmax_s;
+    constexpr int BB0 = (4 * SB > SCRF * 4) ? 4 * SB : SCRF * 4;
+    constexpr int BB = (BB0 + 15) & ~15;
+    __shared__ __align__(16) char smem[BB];
     const int w = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
     const int lane = threadIdx.x & 63;
-    const int pair = w >> 1, p = w & 1;
     const int g4 = lane >> 4, li = lane & 15;
-    char* base = smem + pair * PB;
-    float* scr = (float*)base;
-    const long long eme = (long long)blockIdx.x * 2 + pair;
-    const bool live = eme < nrows;
-    const long long e = live ? (row_order ? row_order[eme] : eme) : 0;
-    long long p0 = 0;
-    int n = 0;
-    if (live) {
-        p0 = indptr[e];
-        n = (int)(indptr[e + 1] - p0);
+    float* scr = (float*)smem;
+    long long e = blockIdx.x;
+    if (e >= nrows) return;
+    if (row_order) e = row_order[e];
+    const long long p0 = indptr[e];
+    const int n = (int)(indptr[e + 1] - p0);
+    if (n == 0) {   // block-uniform: every wave exits together
+        for (int c = threadIdx.x; c < K; c += 256) {
+            out_f32[e * K + c] = 0.0f;
+            if (out_fp8) out_fp8[e * K + c] = 0;
+        }
+        return;
     }
-    const int nch = (n + 31) >> 5;
-    if (threadIdx.x == 0) rmax_s = 0;
-    __syncthreads();
-    if (lane == 0) atomicMax(&rmax_s, (nch + 1) >> 1);
     {   // zero EXT pad rows of my stage buffer
-        char* buf = base + p * SB;
+        char* buf = smem + (long long)w * SB;
         for (int i = lane; i < 14 * 8; i += WAVE) {
             const int row = K + 2 + i / 8, seg = i % 8;
             *(unsigned*)(buf + (long long)row * TROW + seg * 4) = 0u;
         }
     }
-    __syncthreads();
-    const int rmax = rmax_s;
     f32x4 T[NOWN], E[NE];
 #pragma unroll
     for (int t = 0; t < NOWN; ++t) T[t] = f32x4{0, 0, 0, 0};
 #pragma unroll
     for (int t = 0; t < NE; ++t) E[t] = f32x4{0, 0, 0, 0};
-    for (int r0 = 0; r0 < rmax; ++r0) {
-        const int ch = 2 * r0 + p;
-        if (ch < nch)
-            stage_chunk_w<KT, true>(base + p * SB, indices, values, factors,
-                                    p0 + (long long)ch * 32, n - ch * 32,
-                                    lane);
+    const int nch = (n + 31) >> 5;
+    for (int r0 = 0; r0 < nch; r0 += 4) {
+        if (r0 + w < nch)
+            stage_chunk_w<KT, true>(smem + (long long)w * SB, indices,
+                                    values, factors,
+                                    p0 + (long long)(r0 + w) * 32,
+                                    n - (r0 + w) * 32, lane);
         __syncthreads();
-        const int ns = min(2, nch - 2 * r0);
-        for (int sc = 0; sc < ns; ++sc) {   // ns <= 0 pairs skip cleanly
+        const int ns = min(4, nch - r0);
+#pragma unroll 1
+        for (int sc = 0; sc < ns; ++sc) {
             fp8x8 frag[KT + 1];
-            read_frags<KT, true>(base + sc * SB, lane, frag);
-            if (p == 0) wp_mfma<KT, 0>(frag, T, E);
-            else        wp_mfma<KT, 1>(frag, T, E);
+            read_frags<KT, true>(smem + (long long)sc * SB, lane, frag);
+            switch (w) {
+                case 0: wq_mfma<KT, 0>(frag, T, E); break;
+                case 1: wq_mfma<KT, 1>(frag, T, E); break;
+                case 2: wq_mfma<KT, 2>(frag, T, E); break;
+                default: wq_mfma<KT, 3>(frag, T, E); break;
+            }
         }
         __syncthreads();
     }
     // b = hi + lo via scr rows [0,K) hi, [K,2K) lo
-    if (p == 0) wp_dump_b<KT, 0>(E, scr, g4, li);
-    else        wp_dump_b<KT, 1>(E, scr, g4, li);
+    switch (w) {
+        case 0: wq_dump_b<KT, 0>(E, scr, g4, li); break;
+        case 1: wq_dump_b<KT, 1>(E, scr, g4, li); break;
+        case 2: wq_dump_b<KT, 2>(E, scr, g4, li); break;
+        default: wq_dump_b<KT, 3>(E, scr, g4, li); break;
+    }
     __syncthreads();
     float xa = scr[lane] + scr[K + lane];
     float xb = (64 + lane < K) ? scr[64 + lane] + scr[K + 64 + lane] : 0.0f;
     __syncthreads();   // b read before the panel dumps reuse scr
-    if (p == 0) wp_diag<KT, 0>(T, reg * (float)n, g4, li);
-    else        wp_diag<KT, 1>(T, reg * (float)n, g4, li);
     float da = 1.0f, db = 1.0f;
-    if (p == 0) wp_panels<KT, 0>(T, scr, lane, g4, li, p, da, db);
-    else        wp_panels<KT, 1>(T, scr, lane, g4, li, p, da, db);
+    const float regn = reg * (float)n;
+    switch (w) {
+        case 0:
+            wq_diag<KT, 0>(T, regn, g4, li);
+            wq_panels<KT, 0>(T, scr, lane, g4, li, w, da, db);
+            break;
+        case 1:
+            wq_diag<KT, 1>(T, regn, g4, li);
+            wq_panels<KT, 1>(T, scr, lane, g4, li, w, da, db);
+            break;
+        case 2:
+            wq_diag<KT, 2>(T, regn, g4, li);
+            wq_panels<KT, 2>(T, scr, lane, g4, li, w, da, db);
+            break;
+        default:
+            wq_diag<KT, 3>(T, regn, g4, li);
+            wq_panels<KT, 3>(T, scr, lane, g4, li, w, da, db);
+            break;
+    }
     const float ida = da > 0.0f ? 1.0f / da : 0.0f;
     const float idb = db > 0.0f ? 1.0f / db : 0.0f;
-    if (p == 0) wp_forward<KT, 0>(T, scr, xa, xb, ida, idb, lane, g4, li, p);
-    else        wp_forward<KT, 1>(T, scr, xa, xb, ida, idb, lane, g4, li, p);
+    switch (w) {
+        case 0: wq_forward<KT, 0>(T, scr, xa, xb, ida, idb, lane, g4, li); break;
+        case 1: wq_forward<KT, 1>(T, scr, xa, xb, ida, idb, lane, g4, li); break;
+        case 2: wq_forward<KT, 2>(T, scr, xa, xb, ida, idb, lane, g4, li); break;
+        default: wq_forward<KT, 3>(T, scr, xa, xb, ida, idb, lane, g4, li); break;
+    }
     xa *= ida;
     xb *= idb;
-    if (p == 0) wp_backward<KT, 0>(T, scr, xa, xb, ida, idb, lane, g4, li, p);
-    else        wp_backward<KT, 1>(T, scr, xa, xb, ida, idb, lane, g4, li, p);
-    if (live && p == 0) {
+    switch (w) {
+        case 0: wq_backward<KT, 0>(T, scr, xa, xb, ida, idb, lane, g4, li); break;
+        case 1: wq_backward<KT, 1>(T, scr, xa, xb, ida, idb, lane, g4, li); break;
+        case 2: wq_backward<KT, 2>(T, scr, xa, xb, ida, idb, lane, g4, li); break;
+        default: wq_backward<KT, 3>(T, scr, xa, xb, ida, idb, lane, g4, li); break;
+    }
+    if (w == 0) {
         out_f32[e * K + lane] = xa;
         if (out_fp8) out_fp8[e * K + lane] = f2fp8(xa);
         if (64 + lane < K) {
@@ -1344,7 +1361,7 @@ extern "C" hipError_t fma_als_solve_wavefused2(
     const int* row_order, long long nrows, float reg, hipStream_t stream) {
     if (k % 16 || k <= 64 || k > 128 || nrows <= 0)
         return hipErrorInvalidValue;
-    dim3 grid((unsigned)((nrows + 1) / 2)), block(256);
+    dim3 grid((unsigned)nrows), block(256);
     switch (k / 16) {
         case 5: k_als_solve_wavefused2<5><<<grid, block, 0, stream>>>(indptr, indices, values, factors, out_f32, out_fp8, row_order, nrows, reg); break;
         case 6: k_als_solve_wavefused2<6><<<grid, block, 0, stream>>>(indptr, indices, values, factors, out_f32, out_fp8, row_order, nrows, reg); break;
