@@ -5,7 +5,11 @@ Variants: wave-quad fused (wavefused2), block-fused fp8, modular fp8
     python gpu_debug/k128_variants.py [--ratings N] [--users N] [--items N]
 """
 import argparse
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
 import torch
 
