@@ -125,15 +125,7 @@ def als_solve_side(
         o8 = out_fp8 if out_fp8 is not None else _empty(fac.device)
         ro = row_order if row_order is not None else _empty(fac.device)
         korig = other_factors.shape[1]
-        if fp8 and k > 64 and not fused:
-            # 64 < k <= 128 fp8 default: wave-PAIR fused kernel — tiles
-            # split across 2 waves, register LDL, no A round trip and no
-            # Gramian-occupancy-serialized solve
-            out = torch.empty(csr.num_rows, k, dtype=torch.float32,
-                              device=fac.device)
-            ops.als_solve_wavefused2(csr.indptr, csr.indices, csr.values,
-                                     fac, out, o8, ro, float(reg), _stream())
-        elif fused or k > 64:
+        if fused or k > 64:
             # bf16 k>64 (and explicit fused=True): block-fused kernel (no
             # nrows*k*k A round-trip through HBM)
             out = torch.empty(csr.num_rows, k, dtype=torch.float32,

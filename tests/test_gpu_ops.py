@@ -397,8 +397,15 @@ def test_wavefused2_matches_block_fused(gpu, k):
     fac = ops.quantize_fp8(
         torch.randn(250, k, generator=torch.Generator().manual_seed(2))
         * 0.5).to(gpu)
+    import flink_ms_amd._hip_ops as hip
     out8 = torch.empty(csr.num_rows, k, dtype=torch.uint8, device=gpu)
-    out_wp = ops.als_solve_side(csr, fac, reg=0.6, out_fp8=out8)
+    out_wp = torch.empty(csr.num_rows, k, dtype=torch.float32, device=gpu)
+    _e = torch.empty(0, device=gpu)
+    # wave-quad kernel invoked directly (kept as a measured ablation — the
+    # block-fused kernel won the k>64 shootout, gpu_debug/k128_variants.py)
+    hip.als_solve_wavefused2(csr.indptr, csr.indices, csr.values, fac,
+                             out_wp, out8, _e, 0.6,
+                             torch.cuda.current_stream().cuda_stream)
     out_bl = ops.als_solve_side(csr, fac, reg=0.6, fused=True)
     cpu_csr = csr.to("cpu")
     pair_csr = CSR(cpu_csr.indptr, cpu_csr.indices,
@@ -417,7 +424,6 @@ def test_wavefused2_matches_block_fused(gpu, k):
     indptr = torch.tensor([0, 0, csr.nnz, csr.nnz], dtype=torch.int64,
                           device=gpu)
     out3 = torch.empty(3, k, dtype=torch.float32, device=gpu)
-    import flink_ms_amd._hip_ops as hip
     e = torch.empty(0, device=gpu)
     hip.als_solve_wavefused2(indptr, csr.indices, csr.values, fac, out3,
                              e, e, 0.6, torch.cuda.current_stream().cuda_stream)
