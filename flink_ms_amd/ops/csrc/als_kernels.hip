@@ -43,6 +43,11 @@
 // Fused K1+K2: normal equations + Cholesky solve, one entity per block.
 template <int KT>
 __launch_bounds__(256)
+// waves_per_eu(3): without it the allocator splits 93 VGPR + 88 AGPR
+// (181 total -> 2 waves/SIMD); constrained it finds a 127-VGPR, zero-AGPR,
+// zero-spill allocation -> 3-4 waves/SIMD (measured: the K=128 fused
+// kernel is the whole 1B-config iteration)
+__attribute__((amdgpu_waves_per_eu(3)))
 __global__ void k_als_solve_fused(const long long* __restrict__ indptr,
                                   const int* __restrict__ indices,
                                   const float* __restrict__ values,
@@ -110,6 +115,11 @@ __global__ void k_gramian(const long long* __restrict__ indptr,
 // half-iteration's e4m3 factor image alongside fp32 (and optional bf16).
 template <int KT>
 __launch_bounds__(256)
+// waves_per_eu(3): without it the allocator splits 93 VGPR + 88 AGPR
+// (181 total -> 2 waves/SIMD); constrained it finds a 127-VGPR, zero-AGPR,
+// zero-spill allocation -> 3-4 waves/SIMD (measured: the K=128 fused
+// kernel is the whole 1B-config iteration)
+__attribute__((amdgpu_waves_per_eu(3)))
 __global__ void k_als_solve_fused_fp8(const long long* __restrict__ indptr,
                                       const int* __restrict__ indices,
                                       const float* __restrict__ values,
