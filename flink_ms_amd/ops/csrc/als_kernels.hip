@@ -829,6 +829,11 @@ DEV_INLINE void wavefused_diag(f32x4* T, float regn, int g4, int li) {
 
 template <int KT, bool FP8>
 __launch_bounds__(256)
+// waves_per_eu(5): 96 VGPRs with ~12-17 spilled (52-60 B/lane scratch) buys
+// a 5th wave/SIMD over the natural 105-reg allocation — measured 3.23 ->
+// 2.65 ms/iter on the rank-64 headline (the spill traffic hides behind the
+// same gather stalls the extra wave fills)
+__attribute__((amdgpu_waves_per_eu(5)))
 __global__ void k_als_solve_wavefused(const long long* __restrict__ indptr,
                                       const int* __restrict__ indices,
                                       const float* __restrict__ values,
