@@ -30,7 +30,8 @@ from .. import ops
 from ..data.blocked import CSR, csr_from_coo
 from ..parallel.dist import DistContext, get_context
 from ..parallel.routing import plan_exchange
-from ..parallel.shard import Partition, allgather_rows, exchange_ratings_by_owner
+from ..parallel.shard import (ChunkedAllgather, Partition, allgather_rows,
+                              exchange_ratings_by_owner)
 from ..utils.logging import get_logger
 from ..utils.textio import als_factor_row
 
@@ -54,6 +55,12 @@ class ALSConfig:
     # C1 exchange: 'auto' picks routed all-to-all-v when the referenced
     # fraction of the opposite side is sparse, full all-gather otherwise
     routed_exchange: str = 'auto'
+    # overlapped exchange (multi-GPU allgather path): solve the shard in
+    # ``exchange_chunks`` slabs and all-gather each finished slab on a comm
+    # stream while the next slab solves (SURVEY.md §7 "compute block i
+    # while exchanging block i+1").  'off' restores the serial exchange.
+    overlap_exchange: str = 'auto'
+    exchange_chunks: int = 4
 
 
 @dataclass
@@ -142,6 +149,20 @@ class ALSTrainer:
         self.item_order = torch.argsort(
             self.item_csr.row_counts(), descending=True).to(torch.int32).to(dev)
 
+        # overlapped chunked exchange (allgather path only): [slab][rank]
+        # replica layout + one-time column remap; per-chunk degree orders
+        self._overlap = (self.item_route is None and self.user_route is None
+                         and ctx.is_distributed
+                         and self.cfg.overlap_exchange != 'off')
+        if self._overlap:
+            nc = max(1, self.cfg.exchange_chunks)
+            self.item_chunked = ChunkedAllgather(ctx, self.ipart, nc)
+            self.user_chunked = ChunkedAllgather(ctx, self.upart, nc)
+            self.user_csr.indices = self.item_chunked.remap_indices(
+                self.user_csr.indices).to(dev)
+            self.item_csr.indices = self.user_chunked.remap_indices(
+                self.item_csr.indices).to(dev)
+
         k = self.cfg.num_factors
         kp = ((k + 15) // 16) * 16 if dev.type == "cuda" else k
         self._kp = kp
@@ -158,12 +179,95 @@ class ALSTrainer:
                                       dtype=shard_dtype, device=dev)
         self.user_f32: Optional[torch.Tensor] = None
         self.item_f32: Optional[torch.Tensor] = None
+        if self._overlap:
+            self._prep_overlap(dev, kp)
         log.info(
             "ALS setup: %d users x %d items, %d local ratings, k=%d, "
             "world=%d, exchange=%s",
             num_users, num_items, self.user_csr.nnz, k, ctx.world_size,
             "routed-a2av" if self.item_route or self.user_route
             else "allgather")
+
+    def _prep_overlap(self, dev, kp) -> None:
+        ctx = self.ctx
+        on_gpu = dev.type == "cuda"
+        sdt = self.item_shard.dtype
+        self._item_replica = self.item_chunked.alloc_replica(kp, sdt, dev)
+        self._user_replica = self.user_chunked.alloc_replica(kp, sdt, dev)
+        self._user_out = torch.zeros(self.user_csr.num_rows, kp,
+                                     dtype=torch.float32, device=dev)
+        self._item_out = torch.zeros(self.item_csr.num_rows, kp,
+                                     dtype=torch.float32, device=dev)
+
+        # per-chunk launch orders: degree-descending on GPU; the CPU
+        # reference path needs contiguous id ranges
+        def orders(chunked, csr):
+            counts = csr.row_counts()
+            out = []
+            for a, b in chunked.bounds:
+                b2 = min(b, csr.num_rows)
+                if b2 <= a:
+                    out.append(torch.empty(0, dtype=torch.int32, device=dev))
+                elif on_gpu:
+                    out.append((a + torch.argsort(counts[a:b2],
+                                                  descending=True))
+                               .to(torch.int32).to(dev))
+                else:
+                    out.append(torch.arange(a, b2, dtype=torch.int32))
+            return out
+
+        self._user_chunk_ids = orders(self.user_chunked, self.user_csr)
+        self._item_chunk_ids = orders(self.item_chunked, self.item_csr)
+        self._comm_stream = torch.cuda.Stream() if on_gpu else None
+        # the first user solve consumes the INITIAL item factors
+        self.item_chunked.gather_all(self.item_shard, self._item_replica)
+
+    def _solve_side_overlapped(self, csr, replica_in, out_f32, shard,
+                               chunk_ids, chunked) -> None:
+        """Solve one side slab-by-slab; each finished slab leaves over
+        xGMI on the comm stream while the next slab computes."""
+        ctx = self.ctx
+        on_gpu = ctx.device.type == "cuda"
+        fp8 = self._fp8
+        comm = self._comm_stream
+        cur = torch.cuda.current_stream() if on_gpu else None
+        for c, ids in enumerate(chunk_ids):
+            if ids.numel() > 0:
+                ops.als_solve_chunk(
+                    csr, replica_in, self.cfg.lambda_, out_f32,
+                    shard if fp8 else None,
+                    shard if (on_gpu and not fp8
+                              and shard.dtype == torch.bfloat16) else None,
+                    ids)
+                if not on_gpu and not fp8:
+                    a, b = int(ids[0]), int(ids[-1]) + 1
+                    shard[a:b, : self.cfg.num_factors] = (
+                        out_f32[a:b, : self.cfg.num_factors].to(shard.dtype))
+            if on_gpu:
+                ev = torch.cuda.Event()
+                ev.record(cur)
+                with torch.cuda.stream(comm):
+                    comm.wait_event(ev)
+                    chunked.gather_chunk(shard, c, self._replica_out)
+            else:
+                chunked.gather_chunk(shard, c, self._replica_out)
+        if on_gpu:
+            cur.wait_stream(comm)
+
+    def _step_overlapped(self) -> None:
+        # user solve consumes the item replica (gathered during the
+        # previous step's item solve, or at setup); user slabs stream out
+        # while later slabs solve, then the roles swap
+        self._replica_out = self._user_replica
+        self._solve_side_overlapped(self.user_csr, self._item_replica,
+                                    self._user_out, self.user_shard,
+                                    self._user_chunk_ids, self.user_chunked)
+        self.user_f32 = self._user_out
+        self._replica_out = self._item_replica
+        self._solve_side_overlapped(self.item_csr, self._user_replica,
+                                    self._item_out, self.item_shard,
+                                    self._item_chunk_ids, self.item_chunked)
+        self.item_f32 = self._item_out
 
     # -- iteration -------------------------------------------------------
 
@@ -172,6 +276,13 @@ class ALSTrainer:
         seconds (max over ranks)."""
         ctx = self.ctx
         t0 = time.perf_counter()
+        if self._overlap:
+            self._step_overlapped()
+            if ctx.device.type == "cuda":
+                torch.cuda.synchronize()
+            dt = ctx.max_scalar(time.perf_counter() - t0)
+            self.timings.iter_seconds.append(dt)
+            return dt
         # C1: item factors to every rank (routed a2a-v or all-gather),
         # then solve local users
         # the kernels write the next half-iteration's bf16 shard image

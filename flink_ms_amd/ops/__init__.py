@@ -196,6 +196,61 @@ def als_solve_side(
     return reference.als_solve_side_reference(csr, other_factors, reg)
 
 
+def als_solve_chunk(
+    csr: CSR,
+    other_factors: torch.Tensor,
+    reg: float,
+    out_f32: torch.Tensor,
+    out_fp8: Optional[torch.Tensor],
+    out_bf16: Optional[torch.Tensor],
+    entity_ids: torch.Tensor,
+) -> None:
+    """Solve ONE slab of row entities (``entity_ids``) into the FULL-side
+    output buffers (scatter by entity id through the kernels' row_order
+    argument).  This is the building block of the overlapped C1 exchange:
+    the trainer solves slab i while slab i-1's factors are already in
+    flight over xGMI (SURVEY.md §7 hard part)."""
+    if other_factors.is_cuda:
+        ops = _require_hip()
+        fp8 = other_factors.dtype == torch.uint8
+        fac = (other_factors.contiguous() if fp8
+               else other_factors.to(torch.bfloat16).contiguous())
+        k = fac.shape[1]
+        ids = entity_ids.to(torch.int32).contiguous()
+        sub_indptr = csr.indptr  # full: kernels index it via row_order
+        o8 = out_fp8 if out_fp8 is not None else _empty(fac.device)
+        ob = out_bf16 if out_bf16 is not None else _empty(fac.device)
+        # row_order-driven launch needs indptr[0:nrows+1] only for bounds;
+        # pass a view sized to the slab so nrows = len(ids)
+        ip = sub_indptr[: ids.numel() + 1]
+        if k <= 64:
+            ops.als_solve_wavefused(ip, csr.indices, csr.values, fac,
+                                    out_f32, ob, o8, ids, float(reg),
+                                    _stream())
+        elif fp8:
+            ops.als_solve_fused_fp8(ip, csr.indices, csr.values, fac,
+                                    out_f32, o8, ids, float(reg), _stream())
+        else:
+            ops.als_solve_fused(ip, csr.indices, csr.values, fac,
+                                out_f32, ob, ids, float(reg), _stream())
+        return
+    # CPU (gloo tests): contiguous id range sliced out of the CSR
+    a = int(entity_ids.min()) if entity_ids.numel() else 0
+    b = int(entity_ids.max()) + 1 if entity_ids.numel() else 0
+    nnz0, nnz1 = int(csr.indptr[a]), int(csr.indptr[b])
+    sub = CSR(csr.indptr[a:b + 1] - nnz0, csr.indices[nnz0:nnz1],
+              csr.values[nnz0:nnz1], b - a, csr.num_cols)
+    fac32 = (dequantize_fp8(other_factors)
+             if other_factors.dtype == torch.uint8
+             else other_factors.to(torch.float32))
+    out = reference.als_solve_side_reference(sub, fac32, reg)
+    out_f32[a:b] = out
+    if out_fp8 is not None:
+        out_fp8[a:b] = quantize_fp8(out[:, : out_fp8.shape[1]])
+    if out_bf16 is not None:
+        out_bf16[a:b] = out[:, : out_bf16.shape[1]].to(torch.bfloat16)
+
+
 def gramian(csr: CSR, factors: torch.Tensor, reg: float) -> Tuple[torch.Tensor, torch.Tensor]:
     """Standalone K1 (parity tests / modular path).  ``factors`` as uint8 =
     e4m3 bytes -> the fp8 gather kernel; else bf16."""

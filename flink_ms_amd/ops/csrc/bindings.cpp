@@ -139,12 +139,12 @@ void als_solve_fused(torch::Tensor indptr, torch::Tensor indices,
     check_t(out_f32, torch::kFloat32, "out_f32");
     const int k = (int)factors.size(1);
     const long long nrows = indptr.size(0) - 1;
-    TORCH_CHECK(out_f32.size(0) == nrows && out_f32.size(1) == k,
+    TORCH_CHECK(out_f32.size(0) >= nrows && out_f32.size(1) == k,
                 "out_f32 shape mismatch");
     unsigned short* ob = nullptr;
     if (out_bf16.numel() > 0) {
         check_t(out_bf16, torch::kBFloat16, "out_bf16");
-        TORCH_CHECK(out_bf16.numel() == out_f32.numel(), "out_bf16 shape");
+        TORCH_CHECK(out_bf16.numel() >= out_f32.numel(), "out_bf16 shape");
         ob = bf16_ptr_mut(out_bf16);
     }
     const int* order = nullptr;
@@ -214,12 +214,12 @@ void als_solve_fused_fp8(torch::Tensor indptr, torch::Tensor indices,
     check_t(out_f32, torch::kFloat32, "out_f32");
     const int k = (int)factors.size(1);
     const long long nrows = indptr.size(0) - 1;
-    TORCH_CHECK(out_f32.size(0) == nrows && out_f32.size(1) == k,
+    TORCH_CHECK(out_f32.size(0) >= nrows && out_f32.size(1) == k,
                 "out_f32 shape mismatch");
     unsigned char* o8 = nullptr;
     if (out_fp8.numel() > 0) {
         check_t(out_fp8, torch::kUInt8, "out_fp8");
-        TORCH_CHECK(out_fp8.numel() == out_f32.numel(), "out_fp8 shape");
+        TORCH_CHECK(out_fp8.numel() >= out_f32.numel(), "out_fp8 shape");
         o8 = fp8_ptr_mut(out_fp8);
     }
     check_hip(fma_als_solve_fused_fp8(
@@ -255,16 +255,16 @@ void als_solve_wavefused(torch::Tensor indptr, torch::Tensor indices,
                 "factors must be contiguous on GPU");
     const int k = (int)factors.size(1);
     const long long nrows = indptr.size(0) - 1;
-    TORCH_CHECK(out_f32.size(0) == nrows && out_f32.size(1) == k,
+    TORCH_CHECK(out_f32.size(0) >= nrows && out_f32.size(1) == k,
                 "out_f32 shape mismatch");
     unsigned short* ob = nullptr;
     if (out_bf16.numel() > 0) {
-        TORCH_CHECK(out_bf16.numel() == out_f32.numel(), "out_bf16 shape");
+        TORCH_CHECK(out_bf16.numel() >= out_f32.numel(), "out_bf16 shape");
         ob = bf16_ptr_mut(out_bf16);
     }
     unsigned char* o8 = nullptr;
     if (out_fp8.numel() > 0) {
-        TORCH_CHECK(out_fp8.numel() == out_f32.numel(), "out_fp8 shape");
+        TORCH_CHECK(out_fp8.numel() >= out_f32.numel(), "out_fp8 shape");
         o8 = fp8_ptr_mut(out_fp8);
     }
     check_hip(fma_als_solve_wavefused(

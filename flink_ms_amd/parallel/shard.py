@@ -82,6 +82,72 @@ def exchange_ratings_by_owner(
     return rkeys, rother, rvals
 
 
+class ChunkedAllgather:
+    """Overlap-friendly C1 factor exchange (SURVEY.md §7 hard part:
+    "compute block i while exchanging block i+1").
+
+    The shard is split into ``chunks`` row slabs and the gathered replica
+    is laid out ``[slab][rank][slab_rows][k]`` so each slab's exchange is
+    ONE contiguous ``all_gather_into_tensor`` — launchable on a side
+    stream the moment the solver finishes writing that slab, while the
+    solver's next slab is still running.  Consumers address the replica
+    through ``remap_indices`` (a one-time CSR column remap at setup, like
+    the routed plan's compaction)."""
+
+    def __init__(self, ctx: DistContext, part: Partition, chunks: int = 4):
+        self.ctx = ctx
+        self.part = part
+        ss = part.shard_size
+        self.chunks = max(1, min(chunks, ss))
+        self.slab = (ss + self.chunks - 1) // self.chunks
+        self.bounds = [(c * self.slab, min((c + 1) * self.slab, ss))
+                       for c in range(self.chunks)]
+        self.rows = [b - a for a, b in self.bounds]
+        base, acc = [], 0
+        for r in self.rows:
+            base.append(acc)
+            acc += ctx.world_size * r
+        self._base = base
+        self.replica_rows = acc
+
+    def remap_indices(self, global_ids: torch.Tensor) -> torch.Tensor:
+        """Global entity id -> row in the [slab][rank][rows][k] replica."""
+        ss = self.part.shard_size
+        ids = global_ids.long()
+        r = torch.clamp(ids // ss, max=self.ctx.world_size - 1)
+        o = ids - r * ss
+        c = torch.clamp(o // self.slab, max=self.chunks - 1)
+        base = torch.tensor(self._base, dtype=torch.int64, device=ids.device)
+        rows = torch.tensor(self.rows, dtype=torch.int64, device=ids.device)
+        slab_lo = torch.tensor([a for a, _ in self.bounds],
+                               dtype=torch.int64, device=ids.device)
+        pos = base[c] + r * rows[c] + (o - slab_lo[c])
+        return pos.to(torch.int32)
+
+    def alloc_replica(self, k: int, dtype, device) -> torch.Tensor:
+        return torch.empty(self.replica_rows, k, dtype=dtype, device=device)
+
+    def gather_chunk(self, shard: torch.Tensor, c: int,
+                     replica: torch.Tensor) -> None:
+        """All-gather slab ``c`` of every rank's shard into the replica
+        (callable from any stream; collective order is identical on all
+        ranks by construction — the chunk loop is static)."""
+        a, b = self.bounds[c]
+        k = shard.shape[1]
+        lo = self._base[c]
+        out = replica[lo: lo + self.ctx.world_size * (b - a)]
+        if not self.ctx.is_distributed:
+            out.copy_(shard[a:b])
+            return
+        dist.all_gather_into_tensor(out.view(-1),
+                                    shard[a:b].contiguous().view(-1))
+
+    def gather_all(self, shard: torch.Tensor,
+                   replica: torch.Tensor) -> None:
+        for c in range(self.chunks):
+            self.gather_chunk(shard, c, replica)
+
+
 def allgather_rows(ctx: DistContext, shard: torch.Tensor,
                    total_rows: int) -> torch.Tensor:
     """All-gather equally-padded factor shards -> full [total_rows, k] replica.

@@ -238,10 +238,12 @@ def test_distributed_cli_training(tmp_path):
     assert open(out).read().startswith("MEAN,U,")
 
 
-@pytest.mark.timeout(300)
-def test_driver_bench_launch_contract(tmp_path):
+@pytest.mark.timeout(600)
+@pytest.mark.parametrize("world", [2, 8])
+def test_driver_bench_launch_contract(tmp_path, world):
     """The benchmark driver's exact multi-rank launch: torch.distributed.run
-    of bench.py at world_size=2 (gloo/CPU here; RCCL on the GPU node).
+    of bench.py at world_size 2 AND 8 (gloo/CPU here; RCCL on the GPU
+    node — the world-8 case is the r1 VERDICT item-4 readiness drill).
     Rank 0 must print ONE JSON line with the whole-job aggregate."""
     import json
     import subprocess
@@ -249,23 +251,25 @@ def test_driver_bench_launch_contract(tmp_path):
 
     res = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", str(29100 + os.getpid() % 300),
-         "bench.py", "--gpus", "2",
+         "--nproc-per-node", str(world), "--master-addr", "127.0.0.1",
+         "--master-port", str(29100 + (os.getpid() + world) % 300),
+         "bench.py", "--gpus", str(world),
          "--steps", "2", "--warmup", "1", "--device", "cpu",
          "--users-per-gpu", "300", "--items", "200",
          "--ratings-per-gpu", "5000", "--rank", "16",
          "--svm-rows-per-gpu", "400"],
-        capture_output=True, text=True, timeout=240)
+        capture_output=True, text=True, timeout=500)
     assert res.returncode == 0, res.stderr[-2000:]
     json_lines = [ln for ln in res.stdout.splitlines()
                   if ln.startswith("{")]
     assert len(json_lines) == 1, res.stdout
     d = json.loads(json_lines[0])
-    assert d["n_gpus"] == 2 and d["steps"] == 2 and d["scaling"] == "weak"
-    assert d["config"]["global_batch"] == 10_000  # whole-job aggregate
+    assert d["n_gpus"] == world and d["steps"] == 2
+    assert d["scaling"] == "weak"
+    assert d["config"]["global_batch"] == 5000 * world  # whole-job aggregate
     assert d["value"] > 0 and d["ms_per_step"] > 0
-    assert d["config"]["parallelism"] == "dp2+factor-allgather"
+    assert d["config"]["parallelism"] == f"dp{world}+factor-allgather"
+    assert d["config"]["svm_samples_per_sec"] > 0
 
 
 @pytest.mark.timeout(120)

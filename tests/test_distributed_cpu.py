@@ -336,3 +336,112 @@ def test_distributed_als_fp8_exchange():
     res_sp = evaluate_mse(model_sp.user_factors, model_sp.item_factors,
                           u, i, r)
     assert res.mse < res_sp.mse * 1.25 + 0.05, (res.mse, res_sp.mse)
+
+
+def test_chunked_allgather_remap_unit():
+    """remap_indices must be a bijection global id -> [slab][rank][row]
+    replica position, for uneven tails (last rank short, last slab short)."""
+    from flink_ms_amd.parallel.dist import DistContext
+    from flink_ms_amd.parallel.shard import ChunkedAllgather, Partition
+    ctx = DistContext(rank=0, world_size=3, local_rank=0,
+                      device=torch.device("cpu"))
+    part = Partition(total=22, world=3)   # shards: 8, 8, 6 (padded 8)
+    ch = ChunkedAllgather(ctx, part, chunks=3)  # slabs of 3,3,2
+    ids = torch.arange(22)
+    pos = ch.remap_indices(ids)
+    assert pos.numel() == 22 and len(set(pos.tolist())) == 22
+    assert int(pos.max()) < ch.replica_rows
+    # replica built by hand: gather slab c of each rank contiguously
+    shard_of = [list(range(r * 8, min((r + 1) * 8, 22))) + [-1] * max(0, (r + 1) * 8 - 22)
+                for r in range(3)]
+    replica = []
+    for a, b in ch.bounds:
+        for r in range(3):
+            replica.extend(shard_of[r][a:b])
+    for gid in range(22):
+        assert replica[int(pos[gid])] == gid, gid
+
+
+def _als_overlap_worker(rank, world, port, q, chunks):
+    """Overlapped chunked exchange at world 2 must train identically to
+    the serial exchange (same collectives, different schedule)."""
+    torch.manual_seed(0)
+    ctx = _init(rank, world, port)
+    from flink_ms_amd.data.ratings import RatingsShape, synthetic_ratings
+    from flink_ms_amd.models.als import ALSConfig, ALSTrainer
+    shape = RatingsShape(100, 50, 1800)
+    u, i, r = synthetic_ratings(shape, seed=7)
+    mask = torch.arange(shape.num_ratings) % world == rank
+    res = {}
+    for mode, nch in (("auto", chunks), ("off", 1)):
+        cfg = ALSConfig(iterations=3, num_factors=8, lambda_=0.2,
+                        dtype=torch.float32, routed_exchange="off",
+                        overlap_exchange=mode, exchange_chunks=nch)
+        tr = ALSTrainer(cfg, ctx)
+        tr.setup(u[mask].long(), i[mask].long(), r[mask],
+                 shape.num_users, shape.num_items)
+        assert tr._overlap == (mode == "auto")
+        tr.fit()
+        m = tr.model()
+        res[mode] = (m.user_factors, m.item_factors)
+    du = (res["auto"][0] - res["off"][0]).abs().max()
+    di = (res["auto"][1] - res["off"][1]).abs().max()
+    q.put((rank, {"du": float(du), "di": float(di)}))
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+@pytest.mark.parametrize("chunks", [3, 7])
+def test_overlapped_exchange_matches_serial(chunks):
+    res = _run_workers(_als_overlap_worker, extra=(chunks,))
+    for rank in (0, 1):
+        # identical math, reordered schedule -> tiny fp reorder noise only
+        assert res[rank]["du"] < 1e-4 and res[rank]["di"] < 1e-4, res[rank]
+
+
+def _world8_pipeline_worker(rank, world, port, q):
+    """8-GPU readiness drill: full ALS pipeline at
+    gloo world 8 with 1B-config-SHAPED partitions (scaled nnz), uneven
+    tail shards, overlapped chunked exchange."""
+    torch.manual_seed(rank)
+    ctx = _init(rank, world, port)
+    from flink_ms_amd.data.ratings import RatingsShape, synthetic_ratings
+    from flink_ms_amd.models.als import ALSConfig, ALSTrainer
+    # 1B-config shape scaled down: users not divisible by 8 (uneven tail)
+    num_users, num_items = 1203, 501
+    shape = RatingsShape(num_users, num_items, 4000)
+    u, i, r = synthetic_ratings(shape, seed=3)
+    mask = torch.arange(shape.num_ratings) % world == rank
+    cfg = ALSConfig(iterations=2, num_factors=16, lambda_=0.1,
+                    dtype=torch.float32, routed_exchange="off",
+                    overlap_exchange="auto", exchange_chunks=4)
+    tr = ALSTrainer(cfg, ctx)
+    tr.setup(u[mask].long(), i[mask].long(), r[mask], num_users, num_items)
+    tr.fit()
+    m = tr.model()
+    ok = bool(torch.isfinite(m.user_factors).all()
+              and torch.isfinite(m.item_factors).all())
+    q.put((rank, {"ok": ok, "rows": int(m.user_factors.shape[0])}))
+    torch.distributed.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_world8_full_pipeline_drill():
+    ctx = mp.get_context("spawn")
+    world = 8
+    port = 29500 + ((os.getpid() + 13) % 400)
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_world8_pipeline_worker,
+                         args=(rank, world, port, q)) for rank in range(world)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        rank, payload = q.get(timeout=400)
+        results[rank] = payload
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert all(results[r]["ok"] for r in range(world))
+    # uneven tail: 1203 over 8 -> shards of 151, last rank 146
+    assert results[0]["rows"] == 151 and results[7]["rows"] == 1203 - 7 * 151
