@@ -9,6 +9,9 @@
 #include <torch/extension.h>
 
 #include <cstdint>
+#include <cstdlib>
+#include <thread>
+#include <vector>
 
 extern "C" {
 int fma_als_solve_fused(int k, const int64_t* indptr, const int* indices,
@@ -85,6 +88,97 @@ const char* fma_err_str(int err);
 }
 
 namespace {
+
+// ---------------------------------------------------------------- ingest
+// Threaded parser for ALS model-row blocks ("<id>,<U|I>,<f;f;...>\n" x n):
+// the native replacement for the reference's Kafka-consumer row decode
+// (ALSKafkaConsumer.java:73-82).  One pass splits lines across threads;
+// each thread strtoll/strtof-parses its share in place.  Returns
+// (ids int64[n], kinds uint8[n] (0=U,1=I), factors fp32[n][k],
+//  payload byte offsets int64[n], payload byte lengths int64[n], n_bad).
+// Rows with a wrong factor count or malformed fields are marked kind=255
+// and skipped by the caller's scalar fallback.
+py::tuple parse_als_block(py::bytes text_b, int64_t k) {
+    char* base;
+    Py_ssize_t total;
+    PyBytes_AsStringAndSize(text_b.ptr(), &base, &total);
+    // line index (single pass; cheap relative to float parsing)
+    std::vector<std::pair<int64_t, int64_t>> lines;
+    int64_t start = 0;
+    for (int64_t i = 0; i < total; ++i) {
+        if (base[i] == '\n') {
+            if (i > start) lines.emplace_back(start, i);
+            start = i + 1;
+        }
+    }
+    if (start < total) lines.emplace_back(start, total);
+    const int64_t n = (int64_t)lines.size();
+    auto ids = torch::empty({n}, torch::kInt64);
+    auto kinds = torch::empty({n}, torch::kUInt8);
+    auto facs = torch::empty({n, k}, torch::kFloat32);
+    auto poffs = torch::empty({n}, torch::kInt64);
+    auto plens = torch::empty({n}, torch::kInt64);
+    int64_t* idp = ids.data_ptr<int64_t>();
+    uint8_t* kp = kinds.data_ptr<uint8_t>();
+    float* fp = facs.data_ptr<float>();
+    int64_t* op = poffs.data_ptr<int64_t>();
+    int64_t* lp = plens.data_ptr<int64_t>();
+    std::atomic<int64_t> bad{0};
+    const int nthreads = (int)std::min<int64_t>(
+        std::max<int64_t>(1, n / 20000),
+        (int64_t)std::thread::hardware_concurrency());
+    auto work = [&](int64_t lo, int64_t hi) {
+        int64_t my_bad = 0;
+        for (int64_t r = lo; r < hi; ++r) {
+            const char* p = base + lines[r].first;
+            const char* end = base + lines[r].second;
+            kp[r] = 255;
+            char* q;
+            long long id = strtoll(p, &q, 10);
+            if (q == p || q >= end || *q != ',') { ++my_bad; continue; }
+            ++q;
+            uint8_t kind;
+            if (*q == 'U') kind = 0;
+            else if (*q == 'I') kind = 1;
+            else { ++my_bad; continue; }
+            ++q;
+            if (q >= end || *q != ',') { ++my_bad; continue; }
+            ++q;
+            op[r] = q - base;
+            lp[r] = end - q;
+            float* frow = fp + r * k;
+            int64_t c = 0;
+            bool ok = true;
+            while (q < end && c < k) {
+                char* q2;
+                frow[c] = strtof(q, &q2);
+                if (q2 == q) { ok = false; break; }
+                q = q2;
+                ++c;
+                if (q < end) {
+                    if (*q == ';') ++q;
+                    else { ok = false; break; }
+                }
+            }
+            if (!ok || c != k || q < end) { ++my_bad; continue; }
+            idp[r] = (int64_t)id;
+            kp[r] = kind;
+        }
+        bad += my_bad;
+    };
+    if (nthreads <= 1) {
+        work(0, n);
+    } else {
+        std::vector<std::thread> ts;
+        const int64_t per = (n + nthreads - 1) / nthreads;
+        for (int t = 0; t < nthreads; ++t)
+            ts.emplace_back(work, t * per,
+                            std::min<int64_t>(n, (t + 1) * per));
+        for (auto& t : ts) t.join();
+    }
+    return py::make_tuple(ids, kinds, facs, poffs, plens, (int64_t)bad);
+}
+
 
 void check_hip(int err, const char* what) {
     TORCH_CHECK(err == 0, what, " failed: ", fma_err_str(err));
@@ -506,4 +600,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("mfma_probe_f32", &mfma_probe_f32);
     m.def("mfma_probe_bf16", &mfma_probe_bf16);
     m.def("mfma_probe_fp8", &mfma_probe_fp8);
+    m.def("parse_als_block", &parse_als_block);
 }

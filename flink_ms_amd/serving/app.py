@@ -186,11 +186,16 @@ def create_app(als_store: Optional[ALSModelStore] = None,
 
     @app.post("/model/als/load")
     def als_load(body: LoadBody):
+        # bulk path: the whole file block goes through the native threaded
+        # parser + ONE H2D mirror slab (vs row-at-a-time /model/als/rows)
         try:
             rows = _read_rows(body.path)
         except OSError as e:
             raise HTTPException(400, f"cannot read model path: {e}")
-        return _ingest_or_400(als, rows)
+        try:
+            return {"ingested": als.ingest_bulk("\n".join(rows))}
+        except (ValueError, IndexError) as e:
+            raise HTTPException(400, f"malformed model row: {e}")
 
     @app.post("/model/svm/rows")
     def svm_rows(body: RowsBody):

@@ -35,6 +35,96 @@ ALS_STATE_NAME = "ALS_MODEL"   # ALSKafkaConsumer.java:91
 SVM_STATE_NAME = "SVM_MODEL"   # SVMKafkaConsumer.java:91
 
 
+class FactorBlocks:
+    """Growable per-kind factor storage feeding BOTH the in-process
+    train->serve attach and the bulk text ingest: host fp32 rows, a device
+    bf16 mirror for the batched K4/K5 kernels, an int-id -> row map, and a
+    per-row payload source — either a byte slice of the ingested text
+    block (BYTE-EXACT replies) or ``None`` = format lazily in the Java
+    shape.  Re-ingested ids overwrite their idmap entry (last writer wins;
+    superseded rows keep their storage slot)."""
+
+    def __init__(self, device: torch.device):
+        self.device = device
+        self.k: Optional[int] = None
+        self.host: Dict[str, torch.Tensor] = {}
+        self.dev: Dict[str, torch.Tensor] = {}
+        self.length: Dict[str, int] = {"U": 0, "I": 0}
+        self.idmap: Dict[str, Dict[int, int]] = {"U": {}, "I": {}}
+        self.src: Dict[str, List[Optional[Tuple[int, int, int]]]] = {
+            "U": [], "I": []}
+        self.texts: List[bytes] = []
+
+    def _ensure(self, kind: str, add: int, k: int) -> None:
+        if self.k is None:
+            self.k = k
+        need = self.length[kind] + add
+        cur = self.host.get(kind)
+        if cur is None or cur.shape[0] < need:
+            cap = max(1024, need, 2 * (cur.shape[0] if cur is not None else 0))
+            host = torch.zeros(cap, k, dtype=torch.float32)
+            dev = torch.zeros(cap, k, dtype=torch.bfloat16,
+                              device=self.device)
+            if cur is not None:
+                host[: self.length[kind]] = cur[: self.length[kind]]
+                dev[: self.length[kind]] = self.dev[kind][: self.length[kind]]
+            self.host[kind] = host
+            self.dev[kind] = dev
+
+    def add_block(self, kind: str, ids: torch.Tensor, facs: torch.Tensor,
+                  text: Optional[bytes] = None,
+                  offs: Optional[torch.Tensor] = None,
+                  lens: Optional[torch.Tensor] = None) -> None:
+        n = int(ids.numel())
+        if n == 0:
+            return
+        k = int(facs.shape[1])
+        self._ensure(kind, n, k)
+        start = self.length[kind]
+        self.host[kind][start:start + n] = facs.to(torch.float32)
+        # ONE H2D slab (the r1 store copied row at a time)
+        self.dev[kind][start:start + n] = (
+            facs.to(self.device).to(torch.bfloat16))
+        self.idmap[kind].update(zip(ids.tolist(), range(start, start + n)))
+        if text is not None:
+            bi = len(self.texts)
+            self.texts.append(text)
+            self.src[kind].extend(
+                zip([bi] * n, offs.tolist(), lens.tolist()))
+        else:
+            self.src[kind].extend([None] * n)
+        self.length[kind] = start + n
+
+    def row_of(self, kind: str, eid: int) -> int:
+        return self.idmap[kind].get(eid, -1)
+
+    def vector(self, kind: str, eid: int) -> Optional[List[float]]:
+        r = self.row_of(kind, eid)
+        return None if r < 0 else self.host[kind][r].tolist()
+
+    def payload(self, kind: str, eid: int, fmt) -> Optional[str]:
+        r = self.row_of(kind, eid)
+        if r < 0:
+            return None
+        src = self.src[kind][r]
+        if src is not None:
+            bi, off, ln = src
+            return self.texts[bi][off:off + ln].decode("ascii")
+        return fmt(self.host[kind][r].tolist())
+
+    def update_row_(self, kind: str, eid: int, vec: List[float]) -> bool:
+        """In-place overwrite (online-SGD write-back); payload source
+        switches to lazy format."""
+        r = self.row_of(kind, eid)
+        if r < 0:
+            return False
+        t = torch.tensor(vec, dtype=torch.float32)
+        self.host[kind][r] = t
+        self.dev[kind][r] = t.to(self.device).to(torch.bfloat16)
+        self.src[kind][r] = None
+        return True
+
+
 class ALSModelStore:
     """Keyed store ``"<id>-U" / "<id>-I" / "MEAN-U" / "MEAN-I"`` -> factors."""
 
@@ -45,13 +135,12 @@ class ALSModelStore:
         self._payload: Dict[str, str] = {}       # key -> factor string
         self._vec: Dict[str, List[float]] = {}   # parsed cache (fp64 path)
         self._lock = threading.RLock()
-        # device mirror (built lazily; rebuilt on rank change)
+        # device mirror of row-at-a-time ingests (built lazily)
         self._rows: Dict[str, int] = {}          # key -> row in mirror
         self._mirror: Optional[torch.Tensor] = None  # [cap, k] bf16
         self._mirror_len = 0
         self._k: Optional[int] = None
-        self._attached = None      # tensor-direct factors (attach_factors)
-        self._attached_dev = None
+        self._blocks: Optional[FactorBlocks] = None  # attach + bulk ingest
         self._fmt = None
 
     # ------------------------------------------------------------ ingest
@@ -118,36 +207,86 @@ class ALSModelStore:
         from ..utils.textio import format_factors  # local import cycle-safe
         self._fmt = format_factors
         with self._lock:
-            self._attached = {
-                "U": (user_factors.to(torch.float32).cpu(),
-                      {int(v): r for r, v in enumerate(user_ids.tolist())}
-                      if user_ids is not None else None),
-                "I": (item_factors.to(torch.float32).cpu(),
-                      {int(v): r for r, v in enumerate(item_ids.tolist())}
-                      if item_ids is not None else None),
-            }
+            if self._blocks is None:
+                self._blocks = FactorBlocks(self.device)
+            uf = user_factors.to(torch.float32).cpu()
+            itf = item_factors.to(torch.float32).cpu()
+            uid = (user_ids.long() if user_ids is not None
+                   else torch.arange(uf.shape[0]))
+            iid = (item_ids.long() if item_ids is not None
+                   else torch.arange(itf.shape[0]))
+            self._blocks.add_block("U", uid, uf)
+            self._blocks.add_block("I", iid, itf)
             if self._k is None:
                 self._k = int(user_factors.shape[1])
-            # device mirror for the batched paths
-            dev_u = user_factors.to(self.device).to(torch.bfloat16)
-            dev_v = item_factors.to(self.device).to(torch.bfloat16)
-            self._attached_dev = {"U": dev_u, "I": dev_v}
 
     def _attached_row(self, key: str) -> Optional[List[float]]:
-        att = getattr(self, "_attached", None)
-        if att is None:
+        if self._blocks is None:
             return None
         try:
             entity_id, kind = key.rsplit("-", 1)
-            fac, idmap = att.get(kind, (None, None))
-            if fac is None:
-                return None
-            row = idmap[int(entity_id)] if idmap is not None else int(entity_id)
-            if row < 0 or row >= fac.shape[0]:
-                return None
-            return fac[row].tolist()
+            return self._blocks.vector(kind, int(entity_id))
         except (ValueError, KeyError):
             return None
+
+    # ------------------------------------------------------- bulk ingest
+
+    def ingest_bulk(self, text) -> int:
+        """Batched model-row ingest: the native threaded parser
+        (`_hip_ops.parse_als_block`) decodes the whole block in C++, ONE
+        H2D slab fills the device mirror, and query payloads are served as
+        BYTE-EXACT slices of the ingested text (the r1 path did one host
+        alloc + one H2D per row).  Malformed / off-width rows fall back to
+        the scalar path.  ``text`` is str or bytes of newline-separated
+        ``<id>,<U|I>,<f;...>`` rows."""
+        if isinstance(text, str):
+            text = text.encode("ascii", errors="replace")
+        if not text.strip():
+            return 0
+        try:
+            from flink_ms_amd import _hip_ops
+        except Exception:
+            return self.ingest(text.decode("ascii").splitlines())
+        k = self._k
+        if k is None:
+            first = text.split(b"\n", 1)[0]
+            k = first.count(b";") + 1
+        ids, kinds, facs, offs, lens, nbad = _hip_ops.parse_als_block(
+            text, int(k))
+        n = 0
+        with self._lock:
+            if self._blocks is None:
+                from ..utils.textio import format_factors
+                self._fmt = format_factors
+                self._blocks = FactorBlocks(self.device)
+            if self._k is None:
+                self._k = int(k)
+            for kind_code, kind in ((0, "U"), (1, "I")):
+                sel = kinds == kind_code
+                cnt = int(sel.sum())
+                if cnt == 0:
+                    continue
+                self._blocks.add_block(kind, ids[sel], facs[sel], text,
+                                       offs[sel], lens[sel])
+                n += cnt
+            if n:
+                # bulk rows supersede earlier scalar payloads (last-writer
+                # wins) — drop only the stale overlapping entries
+                if self._payload:
+                    for kind_code, kind in ((0, "U"), (1, "I")):
+                        sel = kinds == kind_code
+                        for eid in ids[sel].tolist():
+                            key = f"{eid}-{kind}"
+                            self._payload.pop(key, None)
+                            self._vec.pop(key, None)
+        if int(nbad):
+            # the parser's line list keeps every NON-EMPTY line (match that
+            # exactly so indices align with the kinds array)
+            bad_rows = [r for r in text.decode("ascii").split("\n") if r]
+            rejected = [r for i, r in enumerate(bad_rows)
+                        if i < len(kinds) and int(kinds[i]) == 255]
+            n += self.ingest(rejected)
+        return n
 
     # ------------------------------------------------------------- query
 
@@ -156,13 +295,16 @@ class ALSModelStore:
         QueryClientHelper.java:135-137)."""
         with self._lock:
             payload = self._payload.get(key)
-        if payload is None:
-            vec = self._attached_row(key)
-            if vec is not None:
-                payload = self._fmt(vec)
-                with self._lock:  # cache the lazily formatted payload
+        if payload is None and self._blocks is not None:
+            try:
+                entity_id, kind = key.rsplit("-", 1)
+                payload = self._blocks.payload(kind, int(entity_id),
+                                               self._fmt)
+            except (ValueError, KeyError):
+                payload = None
+            if payload is not None:
+                with self._lock:  # cache the resolved payload
                     self._payload.setdefault(key, payload)
-                    self._vec.setdefault(key, vec)
         return None if payload is None else (key, payload)
 
     def get_vector(self, key: str) -> Optional[List[float]]:
@@ -183,22 +325,20 @@ class ALSModelStore:
 
     def _batch_rows(self, ids: List[str], kind: str
                     ) -> Tuple[torch.Tensor, torch.Tensor]:
-        """Resolve ids -> (mirror_row, attached_row); -1 where absent.
-        Ingested rows (the mirror) take precedence over attached tensors."""
-        att = getattr(self, "_attached", None)
-        idmap = att[kind][1] if att is not None else None
-        att_n = att[kind][0].shape[0] if att is not None else 0
+        """Resolve ids -> (mirror_row, block_row); -1 where absent.
+        Row-at-a-time ingests (the mirror) take precedence over
+        attach/bulk blocks."""
+        blocks = self._blocks
+        idmap = blocks.idmap[kind] if blocks is not None else None
         m_rows, a_rows = [], []
         for s in ids:
             key = als_state_key(s, kind)
             m = self._rows.get(key, -1)
             a = -1
-            if m < 0 and att is not None:
+            if m < 0 and idmap is not None:
                 try:
-                    a = idmap[int(s)] if idmap is not None else int(s)
-                    if a < 0 or a >= att_n:
-                        a = -1
-                except (ValueError, KeyError):
+                    a = idmap.get(int(s), -1)
+                except ValueError:
                     a = -1
             m_rows.append(m)
             a_rows.append(a)
@@ -214,7 +354,7 @@ class ALSModelStore:
             um, ua = self._batch_rows(user_ids, "U")
             im, ia = self._batch_rows(item_ids, "I")
             mirror = self._mirror
-            att_dev = getattr(self, "_attached_dev", None)
+            att_dev = self._blocks.dev if self._blocks is not None else None
         ok = ((um >= 0) | (ua >= 0)) & ((im >= 0) | (ia >= 0))
         preds = torch.zeros(len(user_ids), dtype=torch.float32)
         if int(ok.sum()) == 0:
@@ -306,21 +446,15 @@ class ALSModelStore:
             for key, payload in self._payload.items():
                 entity_id, kind = key.rsplit("-", 1)
                 out.append(f"{entity_id},{kind},{payload}")
-            att = self._attached
-            if att is not None:
+            if self._blocks is not None:
                 for kind in ("U", "I"):
-                    fac, idmap = att.get(kind, (None, None))
-                    if fac is None:
-                        continue
-                    ids = (sorted(idmap.items(), key=lambda kv: kv[1])
-                           if idmap is not None
-                           else [(r, r) for r in range(fac.shape[0])])
-                    for entity_id, row in ids:
-                        key = als_state_key(entity_id, kind)
+                    for eid in self._blocks.idmap[kind]:
+                        key = als_state_key(eid, kind)
                         if key in self._payload:
                             continue
-                        out.append(f"{entity_id},{kind},"
-                                   + self._fmt(fac[row].tolist()))
+                        out.append(f"{eid},{kind},"
+                                   + self._blocks.payload(kind, eid,
+                                                          self._fmt))
             return out
 
     def __len__(self) -> int:

@@ -494,3 +494,62 @@ def test_snapshot_covers_attached_factors():
     assert restored.query("7-I") == store.query("7-I")
     assert restored.predict("5", "2") == pytest.approx(
         store.predict("5", "2"), rel=1e-12)
+
+
+def test_bulk_ingest_parity_and_byte_exact_payloads():
+    """ingest_bulk (native threaded parse + one H2D slab) must agree with
+    the scalar path: same predictions, BYTE-EXACT payload echoes (slices
+    of the ingested text), malformed rows recovered scalar-side, and
+    last-writer-wins vs earlier scalar rows."""
+    rows = ["1,U,0.5;0.25;-1.0", "2,U,1.5;2.5;3.5", "1,I,0.125;0.375;0.75",
+            "7,I,1.0E-8;2.0;4.4028", "MEAN,U,0.1;0.2;0.3"]
+    scalar = ALSModelStore(device=torch.device("cpu"))
+    scalar.ingest(rows)
+    bulk = ALSModelStore(device=torch.device("cpu"))
+    bulk.ingest_row("2,U,9.0;9.0;9.0")  # superseded by the bulk block
+    n = bulk.ingest_bulk("\n".join(rows))
+    assert n == len(rows) - 1 or n == len(rows)  # MEAN row may go scalar
+    for key in ("1-U", "2-U", "1-I", "7-I"):
+        assert bulk.query(key) == scalar.query(key), key
+    assert bulk.predict("1", "1") == pytest.approx(scalar.predict("1", "1"),
+                                                   rel=1e-12)
+    # byte-exact: the quirky exponent form survives verbatim
+    assert bulk.query("7-I")[1] == "1.0E-8;2.0;4.4028"
+    # batched kernel path resolves bulk rows
+    preds, ok = bulk.predict_batch(["1", "2"], ["1", "7"])
+    assert ok.tolist() == [True, True]
+    exp = 0.5 * 0.125 + 0.25 * 0.375 + (-1.0) * 0.75
+    assert preds[0].item() == pytest.approx(exp, abs=0.05)
+    # snapshots cover bulk rows; restore round-trips
+    snap = bulk.snapshot_rows()
+    restored = ALSModelStore(device=torch.device("cpu"))
+    restored.ingest(snap)
+    assert restored.query("7-I") == bulk.query("7-I")
+    # malformed rows: scalar fallback raises the same 400-able error
+    with pytest.raises(ValueError):
+        bulk.ingest_bulk("oops-not-a-row")
+    # MEAN row is queryable (cold-start path)
+    assert bulk.get_vector("MEAN-U") == [0.1, 0.2, 0.3]
+
+
+def test_bulk_ingest_throughput_smoke():
+    """The native parser should clear ~1M rows/s even on this CPU box for
+    small k; the real target is measured on the GPU box
+    (benchmarks/bench_bulk_ingest.py)."""
+    import time as _t
+    k = 16
+    n = 100_000
+    g = torch.Generator().manual_seed(1)
+    fac = torch.randn(n, k, generator=g)
+    lines = [f"{i},U," + ";".join(f"{float(x):.6g}" for x in fac[i][:4])
+             + ";" + ";".join("0.5" for _ in range(k - 4))
+             for i in range(n)]
+    text = "\n".join(lines)
+    store = ALSModelStore(device=torch.device("cpu"))
+    t0 = _t.perf_counter()
+    got = store.ingest_bulk(text)
+    dt = _t.perf_counter() - t0
+    assert got == n
+    rate = n / dt
+    assert rate > 200_000, f"bulk ingest too slow: {rate:.0f} rows/s"
+    assert store.query(f"{n-1}-U") is not None
