@@ -36,13 +36,25 @@ def build_app(params: Params):
         als.ingest(_read_rows(params.get("alsModel")))
     if params.has("svmModel"):
         svm.ingest(_read_rows(params.get("svmModel")))
+    # --stateBackend (ALSKafkaConsumer.java:54-65): 'memory' = snapshots
+    # only; 'fs' = durable WAL + snapshots under --checkpointDataUri (the
+    # Kafka at-least-once parity path); 'rocksdb' is REJECTED — no RocksDB
+    # is linked, and silently mapping it to memory (r1 behavior) hid a
+    # durability downgrade.  Use fs for durability.
     backend = params.get("stateBackend", "memory")
-    if backend not in ("rocksdb", "fs", "memory"):
+    if backend == "rocksdb":
+        raise ValueError(
+            "stateBackend 'rocksdb' is not built into this serving stack; "
+            "use --stateBackend fs (durable WAL + snapshots in "
+            "--checkpointDataUri) or memory")
+    if backend not in ("fs", "memory"):
         raise ValueError(f"unknown stateBackend: {backend}")
     return create_app(
         als, svm,
         checkpoint_data_uri=params.get("checkpointDataUri"),
         checkpoint_interval_ms=params.get_int("checkPointInterval", 60000),
+        state_backend=backend,
+        wal_fsync=params.get_bool("walFsync", False),
     )
 
 

@@ -34,6 +34,7 @@ from fastapi import FastAPI, HTTPException
 from pydantic import BaseModel, Field
 
 from .store import ALSModelStore, SVMModelStore
+from .wal import IngestJournal, journal_dir
 
 
 class RowsBody(BaseModel):
@@ -93,7 +94,15 @@ def _read_rows(path: str) -> List[str]:
 def create_app(als_store: Optional[ALSModelStore] = None,
                svm_store: Optional[SVMModelStore] = None,
                checkpoint_data_uri: Optional[str] = None,
-               checkpoint_interval_ms: int = 60000) -> FastAPI:
+               checkpoint_interval_ms: int = 60000,
+               state_backend: str = "memory",
+               wal_fsync: bool = False) -> FastAPI:
+    """``state_backend``: 'memory' (snapshot-only durability, updates since
+    the last snapshot are lost on crash) or 'fs' (durable: every ingested
+    row is write-ahead journaled under ``<checkpointDataUri>/wal`` before
+    the reply and replayed over the newest snapshot on restart — the
+    Kafka-topic at-least-once contract, ALSKafkaProducer.java:36-37 +
+    ALSKafkaConsumer.java:44-51).  'rocksdb' is rejected at the CLI."""
     app = FastAPI(title="flink_ms_amd model serving")
     # NOTE: explicit None checks — an EMPTY store is falsy (__len__ == 0),
     # so `als_store or ALSModelStore()` would silently drop a store that is
@@ -105,6 +114,20 @@ def create_app(als_store: Optional[ALSModelStore] = None,
     app.state.checkpoint_uri = checkpoint_data_uri
     stop_evt = threading.Event()
     app.state._ckpt_stop = stop_evt
+    if state_backend not in ("memory", "fs"):
+        raise ValueError(f"unsupported state backend: {state_backend}")
+    if state_backend == "fs" and not checkpoint_data_uri:
+        raise ValueError("--stateBackend fs requires --checkpointDataUri")
+    wal = {}
+    if state_backend == "fs":
+        wdir = journal_dir(checkpoint_data_uri)
+        wal = {"als": IngestJournal(wdir, "als", fsync=wal_fsync),
+               "svm": IngestJournal(wdir, "svm", fsync=wal_fsync)}
+    app.state.wal = wal
+
+    def _journal(name: str, rows) -> None:
+        if name in wal:
+            wal[name].append(rows)
 
     # Checkpoint RESTORE (Flink parity: the serving job restores its keyed
     # state from the latest completed checkpoint on restart,
@@ -122,6 +145,16 @@ def create_app(als_store: Optional[ALSModelStore] = None,
             if snaps:
                 with open(os.path.join(checkpoint_data_uri, snaps[-1])) as f:
                     store.ingest([ln for ln in f.read().splitlines() if ln])
+            if name in wal:
+                # at-least-once replay over the snapshot: rows are
+                # last-writer-wins upserts, so re-applying already-covered
+                # rows is idempotent; malformed rows (journaled before
+                # their 400) are skipped
+                for row in wal[name].replay_rows():
+                    try:
+                        store.ingest_row(row)
+                    except (ValueError, IndexError):
+                        pass
 
     def _checkpoint() -> dict:
         if not app.state.checkpoint_uri:
@@ -137,6 +170,9 @@ def create_app(als_store: Optional[ALSModelStore] = None,
                 with open(path, "w") as f:
                     f.write("\n".join(rows) + "\n")
                 n += len(rows)
+            if name in wal and rows:
+                # snapshot now covers every journaled row -> rotate
+                wal[name].rotate()
         return {"written": n, "stamp": stamp}
 
     if checkpoint_data_uri and checkpoint_interval_ms > 0:
@@ -182,6 +218,7 @@ def create_app(als_store: Optional[ALSModelStore] = None,
 
     @app.post("/model/als/rows")
     def als_rows(body: RowsBody):
+        _journal("als", body.rows)
         return _ingest_or_400(als, body.rows)
 
     @app.post("/model/als/load")
@@ -192,6 +229,7 @@ def create_app(als_store: Optional[ALSModelStore] = None,
             rows = _read_rows(body.path)
         except OSError as e:
             raise HTTPException(400, f"cannot read model path: {e}")
+        _journal("als", rows)
         try:
             return {"ingested": als.ingest_bulk("\n".join(rows))}
         except (ValueError, IndexError) as e:
@@ -199,6 +237,7 @@ def create_app(als_store: Optional[ALSModelStore] = None,
 
     @app.post("/model/svm/rows")
     def svm_rows(body: RowsBody):
+        _journal("svm", body.rows)
         return _ingest_or_400(svm, body.rows)
 
     @app.post("/model/svm/load")
@@ -207,6 +246,7 @@ def create_app(als_store: Optional[ALSModelStore] = None,
             rows = _read_rows(body.path)
         except OSError as e:
             raise HTTPException(400, f"cannot read model path: {e}")
+        _journal("svm", rows)
         return _ingest_or_400(svm, rows)
 
     @app.post("/checkpoint")
@@ -274,6 +314,7 @@ def create_app(als_store: Optional[ALSModelStore] = None,
                 if "NaN" in out_row:
                     nan_msgs.append(out_row.split(",", 1)[0])
             emitted.extend(rows)
+        _journal("als", emitted)
         return {"updated": len(emitted) // 2, "rows": emitted,
                 "nan_records": nan_msgs}
 
