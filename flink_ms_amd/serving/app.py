@@ -318,6 +318,34 @@ def create_app(als_store: Optional[ALSModelStore] = None,
         return {"updated": len(emitted) // 2, "rows": emitted,
                 "nan_records": nan_msgs}
 
+    @app.post("/sgd/update_batch")
+    def sgd_update_batch(body: SGDBody):
+        """Batched online SGD through the K4 HIP kernel on the device
+        factor tensors: one launch per source group
+        instead of a python loop of scalar fp64 steps.  Unresolvable
+        ratings keep the scalar MEAN cold-start semantics."""
+        users, items, vals = [], [], []
+        for row in body.ratings:
+            row = row.strip()
+            if not row:
+                continue
+            try:
+                u, i, r = row.split(body.field_delimiter)[:3]
+                vals.append(float(r))
+            except ValueError:
+                raise HTTPException(400, f"malformed rating row: {row!r}")
+            users.append(u)
+            items.append(i)
+        try:
+            batched, scalar, rows = als.sgd_update_batch(
+                users, items, vals, body.learning_rate,
+                body.user_regularization, body.item_regularization)
+        except KeyError as e:
+            raise HTTPException(400, str(e))
+        _journal("als", rows)
+        return {"updated": batched + scalar, "batched": batched,
+                "scalar_fallback": scalar, "rows": rows}
+
     # -------------------------------------------------------------- MSE
 
     @app.post("/mse")

@@ -169,3 +169,39 @@ def range_partition_svm_predict(
         res.latencies_ms.append(ms)
     res.wall_s = time.perf_counter() - t_start
     return res
+
+
+def sgd_update_random(
+        store, num_updates: int = 1000, batch: int = 256,
+        lower_user: int = 0, upper_user: int = 1 << 30,
+        lower_item: int = 0, upper_item: int = 1 << 30,
+        learning_rate: float = 0.1, seed: int = 42,
+        client=None) -> LoadgenResult:
+    """Online-update load generator: random rating
+    triples pushed through the batched K4 path (``sgd_update_batch`` /
+    POST /sgd/update_batch), per-batch latency recorded.  The CSV schema
+    extends the reference harness family (ALSPredictRandom.java:94)."""
+    import random as _random
+    import time as _time
+    rng = _random.Random(seed)
+    res = LoadgenResult()
+    done = 0
+    while done < num_updates:
+        n = min(batch, num_updates - done)
+        users = [str(rng.randint(lower_user, upper_user)) for _ in range(n)]
+        items = [str(rng.randint(lower_item, upper_item)) for _ in range(n)]
+        vals = [rng.random() * 4.5 + 0.5 for _ in range(n)]
+        t0 = _time.perf_counter()
+        if client is not None:
+            r = client.post("/sgd/update_batch", json={
+                "ratings": [f"{u}\t{i}\t{v}" for u, i, v in
+                            zip(users, items, vals)],
+                "learning_rate": learning_rate})
+            r.raise_for_status()
+        else:
+            store.sgd_update_batch(users, items, vals, learning_rate)
+        ms = (_time.perf_counter() - t0) * 1000.0
+        res.rows.append(f"{done},{n},{ms:.3f}")
+        res.millis.append(ms)
+        done += n
+    return res

@@ -426,6 +426,83 @@ class ALSModelStore:
         self.ingest(rows)
         return rows
 
+    def sgd_update_batch(self, user_ids: List[str], item_ids: List[str],
+                         ratings: List[float], learning_rate: float = 0.1,
+                         user_reg: float = 0.0, item_reg: float = 0.0
+                         ) -> Tuple[int, int, List[str]]:
+        """Batched online SGD through the K4 kernel (SGD.java:182-207
+        semantics, v1 simultaneous updates) on the DEVICE factor tensors —
+        the GPU path the reference's per-rating stream never had.  Ratings
+        whose user or item resolves nowhere fall back to the scalar path
+        (MEAN cold-start semantics preserved).  Touched rows are read back
+        once and their payloads re-formatted, so point queries stay
+        coherent (values at bf16 storage precision).  Returns
+        (batched_count, scalar_count, emitted_rows)."""
+        from .. import ops
+        n = len(ratings)
+        with self._lock:
+            um, ua = self._batch_rows(user_ids, "U")
+            im, ia = self._batch_rows(item_ids, "I")
+            mirror = self._mirror
+            blocks = self._blocks
+        ok = ((um >= 0) | (ua >= 0)) & ((im >= 0) | (ia >= 0))
+        emitted: List[str] = []
+        batched = 0
+        r_t = torch.tensor(ratings, dtype=torch.float32)
+        for usrc, umask in (("m", um >= 0), ("a", (um < 0) & (ua >= 0))):
+            for isrc, imask in (("m", im >= 0), ("a", (im < 0) & (ia >= 0))):
+                sel = ok & umask & imask
+                cnt = int(sel.sum())
+                if cnt == 0:
+                    continue
+                U = mirror if usrc == "m" else blocks.dev["U"]
+                V = mirror if isrc == "m" else blocks.dev["I"]
+                su = (um if usrc == "m" else ua)[sel].to(U.device)
+                si = (im if isrc == "m" else ia)[sel].to(V.device)
+                ops.sgd_update(U, V, su, si, r_t[sel].to(U.device),
+                               learning_rate, user_reg, item_reg)
+                batched += cnt
+                # coherence: batched read-back of the touched rows, then
+                # re-format payloads (reference emits updated rows back
+                # through the topic; here the loop closes in-process)
+                new_u = U[su].to(torch.float32).cpu()
+                new_v = V[si].to(torch.float32).cpu()
+                sel_idx = sel.nonzero(as_tuple=True)[0].tolist()
+                with self._lock:
+                    for j, qi in enumerate(sel_idx):
+                        uid, iid = user_ids[qi], item_ids[qi]
+                        uvec = new_u[j].tolist()
+                        ivec = new_v[j].tolist()
+                        urow = (f"{uid},U," + ";".join(
+                            java_double_to_string(x) for x in uvec))
+                        irow = (f"{iid},I," + ";".join(
+                            java_double_to_string(x) for x in ivec))
+                        emitted.extend((urow, irow))
+                        for key, vec, kind, src, row_i in (
+                                (als_state_key(uid, "U"), uvec, "U", usrc,
+                                 int(su[j])),
+                                (als_state_key(iid, "I"), ivec, "I", isrc,
+                                 int(si[j]))):
+                            self._vec[key] = vec
+                            if src == "m":
+                                self._payload[key] = ";".join(
+                                    java_double_to_string(x) for x in vec)
+                            else:
+                                # block row: payload resolves from the
+                                # updated host copy on next query
+                                blocks.host[kind][row_i] = torch.tensor(
+                                    vec, dtype=torch.float32)
+                                blocks.src[kind][row_i] = None
+                                self._payload.pop(key, None)
+        scalar = 0
+        for qi in (~ok).nonzero(as_tuple=True)[0].tolist():
+            emitted.extend(self.sgd_update(
+                user_ids[qi], item_ids[qi], ratings[qi], learning_rate,
+                user_reg, item_reg))
+            scalar += 1
+        assert batched + scalar == n
+        return batched, scalar, emitted
+
     # ------------------------------------------------------- bookkeeping
 
     def keys(self) -> List[str]:

@@ -612,3 +612,63 @@ def test_rocksdb_backend_rejected():
         build_app(Params({"stateBackend": "rocksdb"}))
     with pytest.raises(ValueError, match="checkpointDataUri"):
         build_app(Params({"stateBackend": "fs"}))
+
+
+def test_sgd_update_batch_k4_semantics(als_store):
+    """/sgd/update_batch: K4-kernel batched updates must match the scalar
+    v1 path at storage precision, stay payload-coherent, and fall back to
+    scalar MEAN semantics for unknown ids."""
+    from fastapi.testclient import TestClient
+
+    from flink_ms_amd.serving.app import create_app
+    # two identical stores: one stepped scalar, one batched
+    rows = ["1,U,0.5;0.25", "2,U,1.0;-0.5", "1,I,0.75;0.125",
+            "2,I,0.25;1.0", "MEAN,U,0.1;0.1", "MEAN,I,0.1;0.1"]
+    s_scalar = ALSModelStore(device=torch.device("cpu"))
+    s_scalar.ingest(rows)
+    s_batch = ALSModelStore(device=torch.device("cpu"))
+    s_batch.ingest(rows)
+    ratings = [("1", "1", 4.0), ("2", "2", 1.5), ("77", "1", 3.0)]
+    for u, i, r in ratings:
+        s_scalar.sgd_update(u, i, r, learning_rate=0.1)
+    app = create_app(s_batch, SVMModelStore())
+    with TestClient(app) as c:
+        res = c.post("/sgd/update_batch", json={
+            "ratings": [f"{u}\t{i}\t{r}" for u, i, r in ratings],
+            "learning_rate": 0.1})
+        assert res.status_code == 200
+        d = res.json()
+        assert d["updated"] == 3 and d["scalar_fallback"] == 1  # cold 77
+        for key in ("1-U", "2-U", "1-I", "2-I"):
+            vb = s_batch.get_vector(key)
+            vs = s_scalar.get_vector(key)
+            # batched path stores at bf16 precision
+            assert vb == pytest.approx(vs, rel=2e-2, abs=2e-2), key
+            # payload text coherent with the stored vector
+            pb = s_batch.query(key)[1]
+            assert [float(x) for x in pb.split(";")] == pytest.approx(
+                vb, rel=1e-6), key
+        # cold-start went through MEAN fallback and is now queryable
+        assert s_batch.query("77-U") is not None
+
+
+@pytest.mark.gpu
+def test_sgd_update_batch_gpu_kernel():
+    """K4 on the real device mirror through the store surface (the r1 gap:
+    the batched kernel was only reachable from unit tests)."""
+    store = ALSModelStore(device=torch.device("cuda:0"))
+    store.ingest(["1,U,0.5;0.25;0.5;0.25", "2,I,0.75;0.125;0.25;1.0",
+                  "MEAN,U,0.1;0.1;0.1;0.1", "MEAN,I,0.1;0.1;0.1;0.1"])
+    ref = ALSModelStore(device=torch.device("cpu"))
+    ref.ingest(["1,U,0.5;0.25;0.5;0.25", "2,I,0.75;0.125;0.25;1.0"])
+    ref.sgd_update("1", "2", 4.0, learning_rate=0.1)
+    batched, scalar, rows = store.sgd_update_batch(
+        ["1"], ["2"], [4.0], learning_rate=0.1)
+    assert batched == 1 and scalar == 0 and len(rows) == 2
+    vu = store.get_vector("1-U")
+    assert vu == pytest.approx(ref.get_vector("1-U"), rel=2e-2, abs=2e-2)
+    # bulk-block rows also reachable by the batched kernel
+    store.ingest_bulk("7,U,1.0;1.0;1.0;1.0\n9,I,0.5;0.5;0.5;0.5")
+    b2, s2, _ = store.sgd_update_batch(["7"], ["9"], [2.0])
+    assert b2 == 1 and s2 == 0
+    assert store.get_vector("7-U") != [1.0, 1.0, 1.0, 1.0]
