@@ -245,3 +245,103 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+def kvserver_qps(num_keys=20000, k=64, num_clients=(1, 4, 16),
+                 reqs_per_client=20000):
+    """Native KvState server (serving/csrc/kvserver.cpp) point-query
+    throughput with raw keep-alive sockets (mixed /state lookups and
+    /als/predict dots).  NOTE: the Python clients are GIL-bound; at high
+    client counts this measures the CLIENT ceiling, not the server's."""
+    import socket
+    import threading
+    import time as _t
+
+    from flink_ms_amd import _hip_ops
+    kv = _hip_ops.KvServer()
+    port = kv.start(0)
+    kv.put_rows([f"{i},U," + ";".join(["0.5"] * k)
+                 for i in range(num_keys)]
+                + [f"{i},I," + ";".join(["0.25"] * k)
+                   for i in range(num_keys)])
+    results = {}
+
+    def worker(n, out, idx):
+        s = socket.create_connection(("127.0.0.1", port))
+        s.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+        buf = b""
+        lat = []
+        for i in range(n):
+            path = (f"/state/ALS_MODEL/{i % num_keys}-U" if i % 2 == 0 else
+                    f"/als/predict?user={i % num_keys}"
+                    f"&item={(i * 7) % num_keys}")
+            t0 = _t.perf_counter()
+            s.sendall(f"GET {path} HTTP/1.1\r\nHost: x\r\n\r\n".encode())
+            while b"\r\n\r\n" not in buf:
+                buf += s.recv(65536)
+            head, rest = buf.split(b"\r\n\r\n", 1)
+            cl = int([ln for ln in head.split(b"\r\n")
+                      if ln.lower().startswith(b"content-length")
+                      ][0].split(b":")[1])
+            while len(rest) < cl:
+                rest += s.recv(65536)
+            buf = rest[cl:]
+            lat.append(_t.perf_counter() - t0)
+        out[idx] = lat
+        s.close()
+
+    for nc in num_clients:
+        out = {}
+        n = max(2000, reqs_per_client // nc)
+        ts = [threading.Thread(target=worker, args=(n, out, i))
+              for i in range(nc)]
+        t0 = _t.perf_counter()
+        [t.start() for t in ts]
+        [t.join() for t in ts]
+        wall = _t.perf_counter() - t0
+        lats = sorted(x for v in out.values() for x in v)
+        results[nc] = {
+            "qps": len(lats) / wall,
+            "p50_ms": lats[len(lats) // 2] * 1e3,
+            "p95_ms": lats[int(len(lats) * 0.95)] * 1e3,
+        }
+        print(f"kvserver clients={nc:3d}: {results[nc]['qps']:9.0f} QPS  "
+              f"p50 {results[nc]['p50_ms']:.3f} ms  "
+              f"p95 {results[nc]['p95_ms']:.3f} ms", flush=True)
+    kv.stop()
+    return results
+
+
+def svm_score_latency(num_features=47236, range_size=1000, n_queries=2000,
+                      nnz=50):
+    """SVM classify latency through the store (flat + range-partitioned),
+    the reference's SVMPredictRandom / RangePartitionSVMPredict measure
+    (config 5 promises recommend AND score)."""
+    import random as _r
+    import time as _t
+
+    from flink_ms_amd.serving.store import SVMModelStore
+    rng = _r.Random(5)
+    flat = SVMModelStore()
+    flat.ingest([f"{i},{rng.random():.6f}" for i in range(1, num_features)])
+    ranged = SVMModelStore()
+    rows = {}
+    for i in range(1, num_features):
+        rows.setdefault(i // range_size, []).append(f"{i}:{rng.random():.6f}")
+    ranged.ingest([f"{b}," + ";".join(v) for b, v in rows.items()])
+    out = {}
+    for name, store, rs in (("flat", flat, None),
+                            ("range", ranged, range_size)):
+        lats = []
+        for _ in range(n_queries):
+            pairs = [(str(rng.randint(1, num_features - 1)), rng.random())
+                     for _ in range(nnz)]
+            t0 = _t.perf_counter()
+            store.predict(pairs, range_size=rs)
+            lats.append(_t.perf_counter() - t0)
+        lats.sort()
+        out[name] = {"p50_ms": lats[len(lats) // 2] * 1e3,
+                     "p95_ms": lats[int(len(lats) * 0.95)] * 1e3}
+        print(f"svm {name:6s}: p50 {out[name]['p50_ms']:.4f} ms  "
+              f"p95 {out[name]['p95_ms']:.4f} ms  (nnz={nnz})", flush=True)
+    return out

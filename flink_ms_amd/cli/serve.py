@@ -49,12 +49,24 @@ def build_app(params: Params):
             "--checkpointDataUri) or memory")
     if backend not in ("fs", "memory"):
         raise ValueError(f"unknown stateBackend: {backend}")
+    # --kvPort: start the native C++ KvState query server (Netty
+    # KvStateServer parity) on 127.0.0.1:<port> (0 = ephemeral); the hot
+    # GET surface (/state, /als/predict) answers there off the GIL while
+    # this FastAPI app stays the control plane
+    kv = None
+    if params.has("kvPort"):
+        from flink_ms_amd import _hip_ops
+        kv = _hip_ops.KvServer()
+        bound = kv.start(params.get_int("kvPort", 0))
+        print(f"[kvserver] native KvState server on 127.0.0.1:{bound}",
+              flush=True)
     return create_app(
         als, svm,
         checkpoint_data_uri=params.get("checkpointDataUri"),
         checkpoint_interval_ms=params.get_int("checkPointInterval", 60000),
         state_backend=backend,
         wal_fsync=params.get_bool("walFsync", False),
+        kv_server=kv,
     )
 
 

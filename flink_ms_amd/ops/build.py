@@ -27,6 +27,7 @@ ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
 KERNEL_SOURCES = ["als_kernels.hip", "svm_kernels.hip", "serve_kernels.hip",
                   "debug_kernels.hip"]
+KVSERVER_SRC = "kvserver.cpp"  # serving/csrc: host-only C++ (Netty parity)
 
 
 def _hipcc() -> str:
@@ -56,7 +57,9 @@ def build(verbose: bool = True, force: bool = False) -> Path:
     objdir = PKG_DIR / "ops" / "_build"
     objdir.mkdir(exist_ok=True)
 
-    srcs = [CSRC / s for s in KERNEL_SOURCES] + [CSRC / "bindings.cpp"]
+    kv_src = PKG_DIR / "serving" / "csrc" / KVSERVER_SRC
+    srcs = [CSRC / s for s in KERNEL_SOURCES] + [CSRC / "bindings.cpp",
+                                                 kv_src]
     deps = srcs + [CSRC / "common.hip.h", CSRC / "als_kernels_device.inc",
             Path(__file__)]
     if not force and SO_PATH.exists():
@@ -92,6 +95,17 @@ def build(verbose: bool = True, force: bool = False) -> Path:
         "-x", "c++", "-c", CSRC / "bindings.cpp", "-o", bobj,
     ], verbose)
     objs.append(bobj)
+
+    # native KvState query server (host-only C++, pybind11 only)
+    kobj = objdir / "kvserver.o"
+    _run([
+        _hipcc(), *common,
+        f"-D_GLIBCXX_USE_CXX11_ABI={abi}",
+        *[f"-I{p}" for p in tincs],
+        f"-I{py_inc}",
+        "-x", "c++", "-c", kv_src, "-o", kobj,
+    ], verbose)
+    objs.append(kobj)
 
     _run([
         _hipcc(), "-shared", "-fPIC", *objs, "-o", SO_PATH,

@@ -578,8 +578,11 @@ void dbg_gramian(torch::Tensor indptr, torch::Tensor indices,
 
 }  // namespace
 
+void register_kvserver(pybind11::module_& m);  // serving/csrc/kvserver.cpp
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "flink_ms_amd MI355X (gfx950) HIP kernels";
+    register_kvserver(m);
     m.def("als_solve_fused", &als_solve_fused);
     m.def("gramian", &gramian);
     m.def("gramian_fp8", &gramian_fp8);

@@ -96,7 +96,8 @@ def create_app(als_store: Optional[ALSModelStore] = None,
                checkpoint_data_uri: Optional[str] = None,
                checkpoint_interval_ms: int = 60000,
                state_backend: str = "memory",
-               wal_fsync: bool = False) -> FastAPI:
+               wal_fsync: bool = False,
+               kv_server=None) -> FastAPI:
     """``state_backend``: 'memory' (snapshot-only durability, updates since
     the last snapshot are lost on crash) or 'fs' (durable: every ingested
     row is write-ahead journaled under ``<checkpointDataUri>/wal`` before
@@ -129,6 +130,15 @@ def create_app(als_store: Optional[ALSModelStore] = None,
         if name in wal:
             wal[name].append(rows)
 
+    # native KvState data plane (serving/csrc/kvserver.cpp): a C++ HTTP
+    # server mirroring the ALS keyed state off the GIL; every ALS ingest
+    # is pushed so its answers match this app's byte for byte
+    app.state.kv = kv_server
+
+    def _kv_push(rows) -> None:
+        if kv_server is not None and rows:
+            kv_server.put_rows([r.strip() for r in rows if r.strip()])
+
     # Checkpoint RESTORE (Flink parity: the serving job restores its keyed
     # state from the latest completed checkpoint on restart,
     # ALSKafkaConsumer.java:44-51 enableCheckpointing + restart strategy).
@@ -155,6 +165,9 @@ def create_app(als_store: Optional[ALSModelStore] = None,
                         store.ingest_row(row)
                     except (ValueError, IndexError):
                         pass
+
+    if kv_server is not None and len(als):
+        _kv_push(als.snapshot_rows())
 
     def _checkpoint() -> dict:
         if not app.state.checkpoint_uri:
@@ -219,7 +232,9 @@ def create_app(als_store: Optional[ALSModelStore] = None,
     @app.post("/model/als/rows")
     def als_rows(body: RowsBody):
         _journal("als", body.rows)
-        return _ingest_or_400(als, body.rows)
+        out = _ingest_or_400(als, body.rows)
+        _kv_push(body.rows)
+        return out
 
     @app.post("/model/als/load")
     def als_load(body: LoadBody):
@@ -231,9 +246,11 @@ def create_app(als_store: Optional[ALSModelStore] = None,
             raise HTTPException(400, f"cannot read model path: {e}")
         _journal("als", rows)
         try:
-            return {"ingested": als.ingest_bulk("\n".join(rows))}
+            out = {"ingested": als.ingest_bulk("\n".join(rows))}
         except (ValueError, IndexError) as e:
             raise HTTPException(400, f"malformed model row: {e}")
+        _kv_push(rows)
+        return out
 
     @app.post("/model/svm/rows")
     def svm_rows(body: RowsBody):
@@ -315,6 +332,7 @@ def create_app(als_store: Optional[ALSModelStore] = None,
                     nan_msgs.append(out_row.split(",", 1)[0])
             emitted.extend(rows)
         _journal("als", emitted)
+        _kv_push(emitted)
         return {"updated": len(emitted) // 2, "rows": emitted,
                 "nan_records": nan_msgs}
 
@@ -343,6 +361,7 @@ def create_app(als_store: Optional[ALSModelStore] = None,
         except KeyError as e:
             raise HTTPException(400, str(e))
         _journal("als", rows)
+        _kv_push(rows)
         return {"updated": batched + scalar, "batched": batched,
                 "scalar_fallback": scalar, "rows": rows}
 

@@ -672,3 +672,69 @@ def test_sgd_update_batch_gpu_kernel():
     b2, s2, _ = store.sgd_update_batch(["7"], ["9"], [2.0])
     assert b2 == 1 and s2 == 0
     assert store.get_vector("7-U") != [1.0, 1.0, 1.0, 1.0]
+
+
+def test_kvserver_parity_with_fastapi():
+    """The native KvState server must answer the hot GET surface byte-
+    compatibly with the FastAPI app over the same state, stay in sync
+    through ingest/SGD pushes, and survive concurrent clients."""
+    import json
+    import urllib.request
+
+    from fastapi.testclient import TestClient
+
+    from flink_ms_amd import _hip_ops
+    from flink_ms_amd.serving.app import create_app
+    kv = _hip_ops.KvServer()
+    port = kv.start(0)
+    store = ALSModelStore(device=torch.device("cpu"))
+    app = create_app(store, SVMModelStore(), kv_server=kv)
+    try:
+        with TestClient(app) as c:
+            c.post("/model/als/rows", json={"rows": [
+                "1,U,0.5;0.25", "2,I,0.75;0.125",
+                "MEAN,U,0.1;0.1", "MEAN,I,0.1;0.1"]})
+
+            def kv_get(path):
+                try:
+                    r = urllib.request.urlopen(
+                        f"http://127.0.0.1:{port}{path}")
+                    return r.status, json.loads(r.read())
+                except urllib.error.HTTPError as e:
+                    return e.code, json.loads(e.read())
+
+            for path in ("/state/ALS_MODEL/1-U", "/state/ALS_MODEL/2-I",
+                         "/state/ALS_MODEL/404-U",
+                         "/als/predict?user=1&item=2",
+                         "/als/predict?user=9&item=9"):
+                st, body = kv_get(path)
+                rf = c.get(path)
+                assert st == rf.status_code, path
+                if "predict" in path and body.get("found"):
+                    assert body["prediction"] == pytest.approx(
+                        rf.json()["prediction"], rel=1e-12)
+                    assert body["formatted"] == rf.json()["formatted"]
+                elif st == 200:
+                    assert body == rf.json(), path
+            # SGD updates propagate to the native plane
+            c.post("/sgd/update", json={"ratings": ["1\t2\t4.0"]})
+            st, body = kv_get("/state/ALS_MODEL/1-U")
+            assert body["value"][1] == c.get(
+                "/state/ALS_MODEL/1-U").json()["value"][1]
+            # concurrent clients
+            import threading as th
+            errs = []
+
+            def hammer():
+                try:
+                    for _ in range(200):
+                        s2, b2 = kv_get("/als/predict?user=1&item=2")
+                        assert s2 == 200 and b2["found"]
+                except Exception as e:  # pragma: no cover
+                    errs.append(e)
+            ts = [th.Thread(target=hammer) for _ in range(8)]
+            [t.start() for t in ts]
+            [t.join() for t in ts]
+            assert not errs
+    finally:
+        kv.stop()
