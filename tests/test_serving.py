@@ -669,9 +669,14 @@ def test_sgd_update_batch_gpu_kernel():
     assert vu == pytest.approx(ref.get_vector("1-U"), rel=2e-2, abs=2e-2)
     # bulk-block rows also reachable by the batched kernel
     store.ingest_bulk("7,U,1.0;1.0;1.0;1.0\n9,I,0.5;0.5;0.5;0.5")
-    b2, s2, _ = store.sgd_update_batch(["7"], ["9"], [2.0])
+    # rating 3.0 vs dot 2.0 -> err 1.0 -> a real update
+    b2, s2, _ = store.sgd_update_batch(["7"], ["9"], [3.0])
     assert b2 == 1 and s2 == 0
     assert store.get_vector("7-U") != [1.0, 1.0, 1.0, 1.0]
+    # payload stays coherent with the updated vector
+    p7 = store.query("7-U")[1]
+    assert [float(x) for x in p7.split(";")] == pytest.approx(
+        store.get_vector("7-U"), rel=1e-6)
 
 
 def test_kvserver_parity_with_fastapi():
