@@ -25,14 +25,15 @@ from flink_ms_amd.serving.store import ALSModelStore
 def main(secs: float = 20.0):
     dev = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
     store = ALSModelStore(device=dev)
-    rows = generate_als_model(num_users=20_000, num_items=8_000,
-                              latent_factors=16, seed=3)
+    rows = list(generate_als_model(num_users=20_000, num_items=8_000,
+                                   latent_factors=16, seed=3))
     store.ingest(rows)
     print(f"soak: {len(store)} keys on {dev}", flush=True)
 
     stop = time.time() + secs
     errors = []
-    counts = {"query": 0, "sgd": 0, "ingest": 0, "batch": 0, "ckpt": 0}
+    counts = {"query": 0, "sgd": 0, "ingest": 0, "batch": 0, "ckpt": 0,
+              "sgd_batch": 0, "kv": 0, "bulk": 0}
     lock = threading.Lock()
 
     def bump(k):
@@ -107,9 +108,74 @@ def main(secs: float = 20.0):
                 bump("ckpt")
                 time.sleep(0.5)
 
+    def worker_sgd_batch():
+        rng = random.Random(5)
+        while time.time() < stop:
+            us = [str(rng.randrange(1, 20_001)) for _ in range(128)]
+            its = [str(rng.randrange(1, 8_001)) for _ in range(128)]
+            vals = [rng.uniform(1, 5) for _ in range(128)]
+            try:
+                b, sc, _ = store.sgd_update_batch(us, its, vals,
+                                                  learning_rate=0.001)
+            except Exception as e:  # noqa: BLE001
+                errors.append(f"sgd_batch: {e!r}")
+                return
+            if b + sc != 128:
+                errors.append("sgd_batch count mismatch")
+                return
+            bump("sgd_batch")
+
+    def worker_bulk():
+        rng = random.Random(6)
+        while time.time() < stop:
+            lines = []
+            for _ in range(500):
+                uid = rng.randrange(1, 20_001)
+                vec = ";".join(f"{rng.uniform(-1, 1):.4f}"
+                               for _ in range(16))
+                lines.append(f"{uid},U,{vec}")
+            try:
+                store.ingest_bulk("\n".join(lines))
+            except Exception as e:  # noqa: BLE001
+                errors.append(f"bulk: {e!r}")
+                return
+            bump("bulk")
+            time.sleep(0.05)
+
+    def worker_kv():
+        # native data plane under concurrent updates
+        import json
+        import urllib.request
+        try:
+            from flink_ms_amd import _hip_ops
+            kv = _hip_ops.KvServer()
+            port = kv.start(0)
+        except Exception as e:  # noqa: BLE001
+            errors.append(f"kv start: {e!r}")
+            return
+        rng = random.Random(7)
+        try:
+            kv.put_rows(rows[:5000])
+            while time.time() < stop:
+                kv.put_rows([f"{rng.randrange(1, 5000)},U," + ";".join(
+                    f"{rng.uniform(-1, 1):.4f}" for _ in range(16))])
+                r = urllib.request.urlopen(
+                    f"http://127.0.0.1:{port}/state/ALS_MODEL/"
+                    f"{rng.randrange(1, 5000)}-U", timeout=5)
+                body = json.loads(r.read())
+                if len(body["value"][1].split(";")) != 16:
+                    errors.append("kv bad payload")
+                    return
+                bump("kv")
+        except Exception as e:  # noqa: BLE001
+            errors.append(f"kv: {e!r}")
+        finally:
+            kv.stop()
+
     threads = [threading.Thread(target=t, daemon=True)
                for t in (worker_query, worker_query, worker_sgd,
-                         worker_ingest, worker_batch, worker_ckpt)]
+                         worker_ingest, worker_batch, worker_ckpt,
+                         worker_sgd_batch, worker_bulk, worker_kv)]
     t0 = time.time()
     for t in threads:
         t.start()
