@@ -49,12 +49,35 @@ def server():
     srv.should_exit = True
 
 
-def _free_port():
-    s = socket.socket()
-    s.bind(("127.0.0.1", 0))
-    port = s.getsockname()[1]
-    s.close()
-    return port
+def _free_port(span: int = 1):
+    """Find a port (or a run of ``span`` consecutive free ports — the
+    sharded serve CLI binds base..base+span-1, and probing only base made
+    the shard test flaky under the full suite)."""
+    import random as _r
+    rng = _r.Random()
+    for _ in range(200):
+        socks = []
+        try:
+            s0 = socket.socket()
+            if span == 1:
+                s0.bind(("127.0.0.1", 0))
+                port = s0.getsockname()[1]
+                s0.close()
+                return port
+            port = rng.randint(20000, 55000)
+            s0.bind(("127.0.0.1", port))
+            socks.append(s0)
+            for off in range(1, span):
+                sx = socket.socket()
+                sx.bind(("127.0.0.1", port + off))
+                socks.append(sx)
+            return port
+        except OSError:
+            continue
+        finally:
+            for sx in socks:
+                sx.close()
+    raise RuntimeError("no free port run found")
 
 
 @pytest.mark.timeout(300)
@@ -314,7 +337,7 @@ def test_serve_cli_shards_subprocess(tmp_path):
     import subprocess
     import sys
 
-    base = _free_port()
+    base = _free_port(span=2)
     proc = subprocess.Popen(
         [sys.executable, "-m", "flink_ms_amd.cli.serve", "--shards", "2",
          "--port", str(base), "--device", "cpu"],
