@@ -41,7 +41,7 @@ __global__ void k_sdca_pass(const long long* __restrict__ indptr,
     // (the v gather itself stays at compute time for hogwild freshness).
     struct Meta {
         long long i, e0, e1;
-        float nsq, yi, a, val, vv;
+        float nsq, yi, a, val;
         int idx;
     };
     auto fetch = [&](long long s, Meta& m) {
@@ -55,11 +55,6 @@ __global__ void k_sdca_pass(const long long* __restrict__ indptr,
         const long long t = m.e0 + lane;
         m.idx = (t < m.e1) ? indices[t] : -1;
         m.val = (t < m.e1) ? values[t] : 0.0f;
-        // the first-64-nonzero v gather rides the prefetch too: one
-        // sample of extra staleness, well inside the hogwild contract,
-        // in exchange for removing the dependent v round trip from the
-        // per-sample critical chain (r2: 68% wait-any)
-        m.vv = (m.idx >= 0) ? v[m.idx] : 0.0f;
     };
 
     Meta cur, nxt;
@@ -67,7 +62,7 @@ __global__ void k_sdca_pass(const long long* __restrict__ indptr,
     for (long long s = wave; s < nrows; s = s + nwaves) {
         fetch(s + nwaves, nxt);   // in flight across the current compute
         if (cur.e1 > cur.e0 && cur.nsq != 0.0f) {
-            float part = cur.val * cur.vv;
+            float part = (cur.idx >= 0) ? cur.val * v[cur.idx] : 0.0f;
             for (long long t = cur.e0 + 64 + lane; t < cur.e1; t += WAVE)
                 part += values[t] * v[indices[t]];
             const float dot = wave_reduce_sum(part);
@@ -116,7 +111,7 @@ extern "C" hipError_t fma_sdca_pass(
     // to a synchronous full-batch step and stalls dual convergence
     long long waves = nrows / 16;
     if (waves < 4) waves = 4;
-    if (waves > 16384) waves = 16384;
+    if (waves > 8192) waves = 8192;
     unsigned grid = (unsigned)((waves + 3) / 4);
     k_sdca_pass<<<dim3(grid), dim3(256), 0, stream>>>(
         indptr, indices, values, y, norms_sq, perm, alpha, v, nrows, scale);
