@@ -64,6 +64,10 @@ def parse_args(argv=None):
     p.add_argument("--no-svm", action="store_true",
                    help="skip the CoCoA-SVM secondary bench")
     p.add_argument("--svm-rows-per-gpu", type=int, default=697_641)
+    p.add_argument("--overlap", choices=["auto", "off", "force"],
+                   default="auto",
+                   help="chunked exchange/solve overlap (force = exercise "
+                        "the comm-stream pipeline even at world 1)")
     p.add_argument("--device", default=None, help="cpu override for tests")
     return p.parse_args(argv)
 
@@ -92,7 +96,8 @@ def bench_als(args, ctx):
     cfg = ALSConfig(iterations=args.steps * ips, num_factors=args.rank,
                     lambda_=args.lambda_, seed=args.seed,
                     dtype=torch.bfloat16 if on_gpu else torch.float32,
-                    factor_dtype=args.factor_dtype if on_gpu else "bf16")
+                    factor_dtype=args.factor_dtype if on_gpu else "bf16",
+                    overlap_exchange=args.overlap)
     trainer = ALSTrainer(cfg, ctx)
     trainer.setup(u, i.long(), r, num_users, items)
 

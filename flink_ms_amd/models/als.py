@@ -151,8 +151,12 @@ class ALSTrainer:
 
         # overlapped chunked exchange (allgather path only): [slab][rank]
         # replica layout + one-time column remap; per-chunk degree orders
+        # 'force' enables the chunked machinery at world 1 too (gathers
+        # become device copies) — lets a single-GPU box validate the
+        # stream/event pipeline the 8-GPU run relies on
         self._overlap = (self.item_route is None and self.user_route is None
-                         and ctx.is_distributed
+                         and (ctx.is_distributed
+                              or self.cfg.overlap_exchange == 'force')
                          and self.cfg.overlap_exchange != 'off')
         if self._overlap:
             nc = max(1, self.cfg.exchange_chunks)

@@ -169,3 +169,25 @@ def test_sharded_routing_unit():
         for row in c.rows:
             assert shard_of(als_row_key(row), 4) == sh
     assert sum(len(c.rows) for c in sc.clients) == 50
+
+
+def test_overlap_force_single_process():
+    """overlap_exchange='force' runs the chunked pipeline at world 1
+    (gathers = copies) and must match the plain path."""
+    import torch as T
+
+    from flink_ms_amd.data.ratings import RatingsShape, synthetic_ratings
+    from flink_ms_amd.models.als import ALSConfig, ALSTrainer
+    u, i, r = synthetic_ratings(RatingsShape(150, 70, 2500), seed=9)
+    out = {}
+    for mode in ("off", "force"):
+        tr = ALSTrainer(ALSConfig(iterations=3, num_factors=8, lambda_=0.2,
+                                  dtype=T.float32, overlap_exchange=mode,
+                                  exchange_chunks=3))
+        tr.setup(u.long(), i.long(), r, 150, 70)
+        assert tr._overlap == (mode == "force")
+        tr.fit()
+        m = tr.model()
+        out[mode] = (m.user_factors, m.item_factors)
+    assert (out["off"][0] - out["force"][0]).abs().max() < 1e-4
+    assert (out["off"][1] - out["force"][1]).abs().max() < 1e-4
