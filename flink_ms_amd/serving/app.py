@@ -232,7 +232,15 @@ def create_app(als_store: Optional[ALSModelStore] = None,
     @app.post("/model/als/rows")
     def als_rows(body: RowsBody):
         _journal("als", body.rows)
-        out = _ingest_or_400(als, body.rows)
+        if len(body.rows) >= 256:
+            # producer-sized batches take the bulk path (C++ parse + one
+            # H2D slab); small/interactive batches stay scalar
+            try:
+                out = {"ingested": als.ingest_bulk("\n".join(body.rows))}
+            except (ValueError, IndexError) as e:
+                raise HTTPException(400, f"malformed model row: {e}")
+        else:
+            out = _ingest_or_400(als, body.rows)
         _kv_push(body.rows)
         return out
 

@@ -275,7 +275,7 @@ def test_driver_bench_launch_contract(tmp_path, world):
     res = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", str(world), "--master-addr", "127.0.0.1",
-         "--master-port", str(29100 + (os.getpid() + world) % 300),
+         "--master-port", str(_free_port()),
          "bench.py", "--gpus", str(world),
          "--steps", "2", "--warmup", "1", "--device", "cpu",
          "--users-per-gpu", "300", "--items", "200",
