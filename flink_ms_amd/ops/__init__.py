@@ -144,9 +144,15 @@ def als_solve_side(
             # bound by exactly that A write+read traffic)
             out = torch.empty(csr.num_rows, k, dtype=torch.float32,
                               device=fac.device)
-            ops.als_solve_wavefused(csr.indptr, csr.indices, csr.values,
-                                    fac, out, ob, o8, ro, float(reg),
-                                    _stream())
+            import os as _os
+            if fp8 and _os.environ.get("FMA_WAVEFUSED_DB") == "1":
+                ops.als_solve_wavefused_db(csr.indptr, csr.indices,
+                                           csr.values, fac, out, ob, o8,
+                                           ro, float(reg), _stream())
+            else:
+                ops.als_solve_wavefused(csr.indptr, csr.indices, csr.values,
+                                        fac, out, ob, o8, ro, float(reg),
+                                        _stream())
         else:
             # slab the normal equations: A is nrows*k*k fp32, which at the
             # 1B-rating configs would exceed HBM if materialized whole.

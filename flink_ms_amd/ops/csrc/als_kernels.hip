@@ -827,6 +827,186 @@ DEV_INLINE void wavefused_diag(f32x4* T, float regn, int g4, int li) {
     }
 }
 
+// Split staging for the double-buffered wave-fused variant: issue the
+// gathers for chunk ch+1 (registers only, no wait) BEFORE chunk ch's
+// MFMAs, commit the transpose+LDS writes after — the load latency hides
+// behind the matrix work instead of stalling the wave.
+template <int KT, bool FP8>
+struct StagedRegs {
+    uint4 v[4];
+    unsigned ext_h, ext_l;
+};
+
+template <int KT, bool FP8>
+DEV_INLINE void stage_loads(StagedRegs<KT, FP8>& sr,
+                            const int* __restrict__ indices,
+                            const float* __restrict__ values,
+                            const void* __restrict__ factors_v,
+                            long long base, int nrem, int lane) {
+    constexpr int K = Geo<KT>::K;
+    static_assert(FP8, "double-buffer staging: fp8 path only");
+    const unsigned char* factors = (const unsigned char*)factors_v;
+    constexpr int NT = 8 * (K / 16);
+    if (lane < NT) {
+        const int q = lane & 7, c16 = lane >> 3;
+#pragma unroll
+        for (int m = 0; m < 4; ++m) {
+            uint4 x = {0, 0, 0, 0};
+            const int row = 4 * q + m;
+            if (row < nrem) {
+                const long long col = indices[base + row];
+                x = *(const uint4*)(factors + col * (long long)K + c16 * 16);
+            }
+            sr.v[m] = x;
+        }
+    }
+    if (lane < 8) {
+        unsigned h = 0, l = 0;
+#pragma unroll
+        for (int m = 0; m < 4; ++m) {
+            const int row = 4 * lane + m;
+            const float r = (row < nrem) ? values[base + row] : 0.0f;
+            const unsigned char hb = f2fp8(r);
+            const unsigned char lb = f2fp8(r - fp82f(hb));
+            h |= ((unsigned)hb) << (8 * m);
+            l |= ((unsigned)lb) << (8 * m);
+        }
+        sr.ext_h = h;
+        sr.ext_l = l;
+    }
+}
+
+template <int KT, bool FP8>
+DEV_INLINE void stage_commit(const StagedRegs<KT, FP8>& sr, char* buf,
+                             int lane) {
+    constexpr int K = Geo<KT>::K;
+    constexpr int TROW = Geo<KT>::TROW8;
+    constexpr int NT = 8 * (K / 16);
+    if (lane < NT) {
+        const int q = lane & 7, c16 = lane >> 3;
+        char* rb = buf + (long long)(16 * c16) * TROW + q * 4;
+#pragma unroll
+        for (int cc = 0; cc < 16; ++cc) {
+            unsigned wd = 0;
+#pragma unroll
+            for (int m = 0; m < 4; ++m)
+                wd |= (((&sr.v[m].x)[cc >> 2] >> (8 * (cc & 3))) & 0xffu)
+                      << (8 * m);
+            *(unsigned*)(rb + (long long)cc * TROW) = wd;
+        }
+    }
+    if (lane < 8) {
+        *(unsigned*)(buf + (long long)K * TROW + lane * 4) = sr.ext_h;
+        *(unsigned*)(buf + (long long)(K + 1) * TROW + lane * 4) = sr.ext_l;
+    }
+}
+
+// Double-buffered wave-fused ALS solve (fp8, k <= 64): same math as
+// k_als_solve_wavefused, software-pipelined staging.
+template <int KT>
+__launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(4)))
+__global__ void k_als_solve_wavefused_db(const long long* __restrict__ indptr,
+                                         const int* __restrict__ indices,
+                                         const float* __restrict__ values,
+                                         const void* __restrict__ factors,
+                                         float* __restrict__ out_f32,
+                                         unsigned short* __restrict__ out_bf16,
+                                         unsigned char* __restrict__ out_fp8,
+                                         const int* __restrict__ row_order,
+                                         long long nrows, float reg) {
+    constexpr int K = KT * 16;
+    static_assert(K <= 64, "wave-fused path handles k <= 64");
+    constexpr int NA = KT * (KT + 1) / 2;
+    constexpr int TROW = Geo<KT>::TROW8;
+    constexpr int SB = (K + 16) * TROW;
+    constexpr int SCRF = (K * 17 > 1040) ? K * 17 : 1040;
+    constexpr int WB0 = (2 * SB > SCRF * 4) ? 2 * SB : SCRF * 4;
+    constexpr int WB = (WB0 + 15) & ~15;
+    __shared__ __align__(16) char smem[4 * WB];
+    const int w = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
+    const int lane = threadIdx.x & 63;
+    const long long e0 = (long long)blockIdx.x * 4 + w;
+    if (e0 >= nrows) return;
+    const long long e = row_order ? row_order[e0] : e0;
+    char* base = smem + (long long)w * WB;
+    float* scr = (float*)base;
+    const int g4 = lane >> 4, li = lane & 15;
+    const long long p0 = indptr[e];
+    const int n = (int)(indptr[e + 1] - p0);
+    if (n == 0) {
+        if (lane < K) {
+            out_f32[e * K + lane] = 0.0f;
+            if (out_bf16) out_bf16[e * K + lane] = 0;
+            if (out_fp8) out_fp8[e * K + lane] = 0;
+        }
+        return;
+    }
+    for (int b = 0; b < 2; ++b)
+        for (int i = lane; i < 14 * 8; i += WAVE) {
+            const int row = K + 2 + i / 8, seg = i % 8;
+            *(unsigned*)(base + b * SB + (long long)row * TROW + seg * 4) =
+                0u;
+        }
+    f32x4 T[NA], E[KT];
+#pragma unroll
+    for (int t = 0; t < NA; ++t) T[t] = f32x4{0, 0, 0, 0};
+#pragma unroll
+    for (int p = 0; p < KT; ++p) E[p] = f32x4{0, 0, 0, 0};
+    const int nchunks = (n + 31) >> 5;
+    StagedRegs<KT, true> sr;
+    stage_loads<KT, true>(sr, indices, values, factors, p0, n, lane);
+    stage_commit<KT, true>(sr, base, lane);
+#pragma unroll 1
+    for (int ch = 0; ch < nchunks; ++ch) {
+        if (ch + 1 < nchunks)
+            stage_loads<KT, true>(sr, indices, values, factors,
+                                  p0 + (long long)(ch + 1) * 32,
+                                  n - (ch + 1) * 32, lane);
+        WREG_FENCE();
+        typename FragT<true>::type frag[KT + 1];
+        read_frags<KT, true>(base + (ch & 1) * SB, lane, frag);
+        WREG_FENCE();
+        mfma_lower_tiles<KT, true>(frag, T);
+        mfma_ext_tiles<KT, true>(frag, E);
+        if (ch + 1 < nchunks)
+            stage_commit<KT, true>(sr, base + ((ch + 1) & 1) * SB, lane);
+    }
+    wavefused_dump_b<KT>(E, scr, g4, li);
+    WREG_FENCE();
+    float x0 = (lane < K) ? scr[lane] + scr[64 + lane] : 0.0f;
+    WREG_FENCE();
+    wavefused_diag<KT>(T, reg * (float)n, g4, li);
+    float d0 = 1.0f;
+    wreg_panels<KT, 0>(T, scr, lane, g4, li, d0);
+    const float id0 = d0 > 0.0f ? 1.0f / d0 : 0.0f;
+    wreg_forward<KT, 0>(T, scr, x0, id0, lane, g4, li);
+    x0 *= id0;
+    wreg_backward<KT, KT - 1>(T, scr, x0, id0, lane, g4, li);
+    if (lane < K) {
+        out_f32[e * K + lane] = x0;
+        if (out_bf16) out_bf16[e * K + lane] = f2bf(x0);
+        if (out_fp8) out_fp8[e * K + lane] = f2fp8(x0);
+    }
+}
+
+extern "C" hipError_t fma_als_solve_wavefused_db(
+    int k, const long long* indptr, const int* indices, const float* values,
+    const void* factors, float* out_f32, unsigned short* out_bf16,
+    unsigned char* out_fp8, const int* row_order, long long nrows,
+    float reg, hipStream_t stream) {
+    if (k % 16 || k < 16 || k > 64 || nrows <= 0) return hipErrorInvalidValue;
+    dim3 grid((unsigned)((nrows + 3) / 4)), block(256);
+    switch (k / 16) {
+        case 1: k_als_solve_wavefused_db<1><<<grid, block, 0, stream>>>(indptr, indices, values, factors, out_f32, out_bf16, out_fp8, row_order, nrows, reg); break;
+        case 2: k_als_solve_wavefused_db<2><<<grid, block, 0, stream>>>(indptr, indices, values, factors, out_f32, out_bf16, out_fp8, row_order, nrows, reg); break;
+        case 3: k_als_solve_wavefused_db<3><<<grid, block, 0, stream>>>(indptr, indices, values, factors, out_f32, out_bf16, out_fp8, row_order, nrows, reg); break;
+        case 4: k_als_solve_wavefused_db<4><<<grid, block, 0, stream>>>(indptr, indices, values, factors, out_f32, out_bf16, out_fp8, row_order, nrows, reg); break;
+        default: return hipErrorInvalidValue;
+    }
+    return hipGetLastError();
+}
+
 template <int KT, bool FP8>
 __launch_bounds__(256)
 // waves_per_eu(5): 96 VGPRs with ~12-17 spilled (52-60 B/lane scratch) buys

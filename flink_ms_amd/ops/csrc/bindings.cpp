@@ -51,6 +51,13 @@ int fma_sgd_update(unsigned short* U, unsigned short* V,
                    const int64_t* u_idx, const int64_t* i_idx,
                    const float* r, float* err_out, int64_t nq, int k,
                    float lr, float user_reg, float item_reg, void* stream);
+int fma_als_solve_wavefused_db(int k, const int64_t* indptr,
+                               const int* indices, const float* values,
+                               const void* factors, float* out_f32,
+                               unsigned short* out_bf16,
+                               unsigned char* out_fp8,
+                               const int* row_order, int64_t nrows,
+                               float reg, void* stream);
 int fma_als_solve_wavefused(int k, int fp8, const int64_t* indptr,
                             const int* indices, const float* values,
                             const void* factors, float* out_f32,
@@ -370,6 +377,34 @@ void als_solve_wavefused(torch::Tensor indptr, torch::Tensor indices,
               "als_solve_wavefused");
 }
 
+// double-buffered variant (fp8 only): software-pipelined staging
+void als_solve_wavefused_db(torch::Tensor indptr, torch::Tensor indices,
+                            torch::Tensor values, torch::Tensor factors,
+                            torch::Tensor out_f32, torch::Tensor out_bf16,
+                            torch::Tensor out_fp8, torch::Tensor row_order,
+                            double reg, int64_t stream) {
+    check_t(indptr, torch::kInt64, "indptr");
+    check_t(indices, torch::kInt32, "indices");
+    check_t(values, torch::kFloat32, "values");
+    check_t(factors, torch::kUInt8, "factors");
+    check_t(out_f32, torch::kFloat32, "out_f32");
+    const int k = (int)factors.size(1);
+    const long long nrows = indptr.size(0) - 1;
+    TORCH_CHECK(out_f32.size(0) >= nrows && out_f32.size(1) == k,
+                "out_f32 shape mismatch");
+    unsigned short* ob = nullptr;
+    if (out_bf16.numel() > 0) ob = bf16_ptr_mut(out_bf16);
+    unsigned char* o8 = nullptr;
+    if (out_fp8.numel() > 0) o8 = fp8_ptr_mut(out_fp8);
+    check_hip(fma_als_solve_wavefused_db(
+                  k, indptr.data_ptr<int64_t>(), indices.data_ptr<int>(),
+                  values.data_ptr<float>(), factors.data_ptr(),
+                  out_f32.data_ptr<float>(), ob, o8,
+                  order_ptr(row_order, nrows), nrows, (float)reg,
+                  (void*)stream),
+              "als_solve_wavefused_db");
+}
+
 // wave-pair fused path for 64 < k <= 128 (fp8 gathers only)
 void als_solve_wavefused2(torch::Tensor indptr, torch::Tensor indices,
                           torch::Tensor values, torch::Tensor factors,
@@ -589,6 +624,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("als_solve_fused_fp8", &als_solve_fused_fp8);
     m.def("als_solve_wavefused", &als_solve_wavefused);
     m.def("als_solve_wavefused2", &als_solve_wavefused2);
+    m.def("als_solve_wavefused_db", &als_solve_wavefused_db);
     m.def("cholesky_solve", &cholesky_solve);
     m.def("cholesky_solve_ph", &cholesky_solve_ph);
     m.def("ldl_solve_wave", &ldl_solve_wave);
