@@ -24,3 +24,9 @@ Layout:
 """
 
 __version__ = "0.1.0"
+
+# _hip_ops links libtorch: importing it before torch's python runtime is
+# initialized segfaults in tensor allocation.  Importing torch here makes
+# `from flink_ms_amd import _hip_ops` safe in any order (observed with the
+# standalone KvServer / parser probes).
+import torch  # noqa: E402,F401

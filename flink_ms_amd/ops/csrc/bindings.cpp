@@ -8,6 +8,7 @@
 
 #include <torch/extension.h>
 
+#include <charconv>
 #include <cstdint>
 #include <cstdlib>
 #include <thread>
@@ -105,10 +106,15 @@ namespace {
 //  payload byte offsets int64[n], payload byte lengths int64[n], n_bad).
 // Rows with a wrong factor count or malformed fields are marked kind=255
 // and skipped by the caller's scalar fallback.
-py::tuple parse_als_block(py::bytes text_b, int64_t k) {
-    char* base;
-    Py_ssize_t total;
-    PyBytes_AsStringAndSize(text_b.ptr(), &base, &total);
+py::tuple parse_als_block(py::object text_o, int64_t k) {
+    // raw CPython buffer protocol: accepts bytes AND mmap objects
+    // (zero-copy parse of a disk-resident model file — the
+    // larger-than-memory serving path)
+    Py_buffer view;
+    if (PyObject_GetBuffer(text_o.ptr(), &view, PyBUF_SIMPLE) != 0)
+        throw py::error_already_set();
+    char* base = (char*)view.buf;
+    Py_ssize_t total = view.len;
     // line index (single pass; cheap relative to float parsing)
     std::vector<std::pair<int64_t, int64_t>> lines;
     int64_t start = 0;
@@ -137,13 +143,18 @@ py::tuple parse_als_block(py::bytes text_b, int64_t k) {
     auto work = [&](int64_t lo, int64_t hi) {
         int64_t my_bad = 0;
         for (int64_t r = lo; r < hi; ++r) {
-            const char* p = base + lines[r].first;
+            // std::from_chars everywhere: bounded by `end` (the input may
+            // be an mmap with no trailing NUL) and faster than strtod
+            const char* q = base + lines[r].first;
             const char* end = base + lines[r].second;
             kp[r] = 255;
-            char* q;
-            long long id = strtoll(p, &q, 10);
-            if (q == p || q >= end || *q != ',') { ++my_bad; continue; }
-            ++q;
+            long long id;
+            auto ir = std::from_chars(q, end, id, 10);
+            if (ir.ec != std::errc() || ir.ptr >= end || *ir.ptr != ',') {
+                ++my_bad;
+                continue;
+            }
+            q = ir.ptr + 1;
             uint8_t kind;
             if (*q == 'U') kind = 0;
             else if (*q == 'I') kind = 1;
@@ -157,10 +168,10 @@ py::tuple parse_als_block(py::bytes text_b, int64_t k) {
             int64_t c = 0;
             bool ok = true;
             while (q < end && c < k) {
-                char* q2;
-                frow[c] = strtof(q, &q2);
-                if (q2 == q) { ok = false; break; }
-                q = q2;
+                auto fr = std::from_chars(q, end, frow[c],
+                                          std::chars_format::general);
+                if (fr.ec != std::errc()) { ok = false; break; }
+                q = fr.ptr;
                 ++c;
                 if (q < end) {
                     if (*q == ';') ++q;
@@ -183,6 +194,7 @@ py::tuple parse_als_block(py::bytes text_b, int64_t k) {
                             std::min<int64_t>(n, (t + 1) * per));
         for (auto& t : ts) t.join();
     }
+    PyBuffer_Release(&view);
     return py::make_tuple(ids, kinds, facs, poffs, plens, (int64_t)bad);
 }
 
