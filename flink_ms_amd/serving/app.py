@@ -43,6 +43,7 @@ class RowsBody(BaseModel):
 
 class LoadBody(BaseModel):
     path: str
+    spill: bool = False   # ALS only: mmap the file, keep factors off-host
 
 
 class SVMPredictBody(BaseModel):
@@ -247,7 +248,17 @@ def create_app(als_store: Optional[ALSModelStore] = None,
     @app.post("/model/als/load")
     def als_load(body: LoadBody):
         # bulk path: the whole file block goes through the native threaded
-        # parser + ONE H2D mirror slab (vs row-at-a-time /model/als/rows)
+        # parser + ONE H2D mirror slab (vs row-at-a-time /model/als/rows).
+        # spill=true (single file): mmap-backed larger-than-memory load —
+        # factors live in the device mirror + byte slices of the file.
+        if body.spill and os.path.isfile(body.path):
+            try:
+                n = als.ingest_bulk_file(body.path)
+            except OSError as e:
+                raise HTTPException(400, f"cannot read model path: {e}")
+            except (ValueError, IndexError) as e:
+                raise HTTPException(400, f"malformed model row: {e}")
+            return {"ingested": n, "spill": True}
         try:
             rows = _read_rows(body.path)
         except OSError as e:

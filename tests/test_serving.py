@@ -743,3 +743,40 @@ def test_kvserver_parity_with_fastapi():
             assert not errs
     finally:
         kv.stop()
+
+
+def test_ingest_bulk_file_spill_mode(tmp_path):
+    """Larger-than-memory serving: mmap-backed bulk load keeps factors
+    only in the device mirror + on-disk byte slices (no dense host copy),
+    while queries, predicts, batched kernels, SGD write-backs and
+    snapshots all stay correct."""
+    rows = [f"{i},U,0.5;0.25;{i % 7}.5" for i in range(200)]
+    rows += [f"{i},I,1.0;0.125;0.75" for i in range(100)]
+    path = tmp_path / "model.txt"
+    path.write_text("\n".join(rows) + "\n")
+    store = ALSModelStore(device=torch.device("cpu"))
+    n = store.ingest_bulk_file(str(path))
+    assert n == 300
+    blocks = store._blocks
+    assert "U" not in blocks.host and "I" not in blocks.host  # spilled
+    # byte-exact payloads from the mapped file
+    assert store.query("13-U")[1] == "0.5;0.25;6.5"
+    assert store.get_vector("13-U") == [0.5, 0.25, 6.5]
+    p0 = store.predict("3", "4")
+    assert p0 == pytest.approx(0.5 * 1.0 + 0.25 * 0.125 + 3.5 * 0.75)
+    preds, ok = store.predict_batch(["3"], ["4"])
+    assert bool(ok[0]) and preds[0].item() == pytest.approx(p0, rel=2e-2)
+    # SGD write-back lands in the overlay, queries stay coherent
+    b, sc, _ = store.sgd_update_batch(["3"], ["4"], [5.0],
+                                      learning_rate=0.1)
+    assert b == 1
+    v = store.get_vector("3-U")
+    assert v != [0.5, 0.25, 3.5]
+    assert [float(x) for x in store.query("3-U")[1].split(";")] == \
+        pytest.approx(v, rel=1e-6)
+    # snapshot covers all rows; restore round-trips
+    snap = store.snapshot_rows()
+    assert len(snap) == 300
+    restored = ALSModelStore(device=torch.device("cpu"))
+    restored.ingest(snap)
+    assert restored.query("13-U") == store.query("13-U")
