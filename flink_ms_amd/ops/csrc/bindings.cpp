@@ -168,9 +168,18 @@ py::tuple parse_als_block(py::object text_o, int64_t k) {
             int64_t c = 0;
             bool ok = true;
             while (q < end && c < k) {
-                auto fr = std::from_chars(q, end, frow[c],
+                // parse as double then narrow: float-subnormal payloads
+                // (e.g. 4.9e-324) must flush like strtof, not reject
+                double dv;
+                auto fr = std::from_chars(q, end, dv,
                                           std::chars_format::general);
-                if (fr.ec != std::errc()) { ok = false; break; }
+                if (fr.ec == std::errc::result_out_of_range) {
+                    dv = 0.0;  // double-range underflow only
+                } else if (fr.ec != std::errc()) {
+                    ok = false;
+                    break;
+                }
+                frow[c] = (float)dv;
                 q = fr.ptr;
                 ++c;
                 if (q < end) {

@@ -56,6 +56,7 @@ class FactorBlocks:
             "U": [], "I": []}
         self.texts: list = []                   # bytes or mmap objects
         self._overlay: Dict[Tuple[str, int], List[float]] = {}
+        self.spilled: Dict[str, bool] = {"U": False, "I": False}
 
     def _ensure(self, kind: str, add: int, k: int) -> None:
         if self.k is None:
@@ -69,8 +70,10 @@ class FactorBlocks:
             if cur is not None:
                 dev[: self.length[kind]] = cur[: self.length[kind]]
             self.dev[kind] = dev
-            old_host = self.host.get(kind)
-            if old_host is not None or kind not in self.host:
+            # a spilled kind STAYS spilled: re-creating a host tensor here
+            # would hand out zeros for the earlier disk-backed rows
+            if not self.spilled[kind]:
+                old_host = self.host.get(kind)
                 host = torch.zeros(cap, k, dtype=torch.float32)
                 if old_host is not None:
                     host[: self.length[kind]] = old_host[: self.length[kind]]
@@ -92,8 +95,11 @@ class FactorBlocks:
         k = int(facs.shape[1])
         self._ensure(kind, n, k)
         start = self.length[kind]
-        if not keep_host:
-            # drop the host side for this kind entirely (spill mode)
+        if not keep_host and not self.spilled[kind]:
+            # drop the host side for this kind entirely (spill mode);
+            # later keep_host blocks for this kind stay device+lazy too
+            # (their vector() reads fall to the bf16 device row)
+            self.spilled[kind] = True
             self.host.pop(kind, None)
         host = self.host.get(kind)
         if host is not None:
