@@ -15,9 +15,15 @@ import torch
 WORLD = 2
 
 
+_PORT_SALT = [0]
+
+
 def _run_workers(target, extra=()):
     ctx = mp.get_context("spawn")
-    port = 29500 + (os.getpid() % 500)
+    # unique port per CALL: re-binding the same pid-derived port across
+    # sequential tests in one session flakes on TIME_WAIT
+    _PORT_SALT[0] += 1
+    port = 20000 + ((os.getpid() * 13 + _PORT_SALT[0] * 101) % 20000)
     q = ctx.Queue()
     procs = [ctx.Process(target=target, args=(rank, WORLD, port, q, *extra))
              for rank in range(WORLD)]
@@ -147,7 +153,8 @@ def test_distributed_als_world4_routed():
     """4-rank gloo run with the routed all-to-all-v exchange forced on:
     wider routing topology than the world-2 cases."""
     ctx = mp.get_context("spawn")
-    port = 29500 + ((os.getpid() + 7) % 500)
+    _PORT_SALT[0] += 1
+    port = 20000 + ((os.getpid() * 13 + _PORT_SALT[0] * 101) % 20000)
     q = ctx.Queue()
     procs = [ctx.Process(target=_als_worker, args=(rank, 4, port, q, "on"))
              for rank in range(4)]
@@ -429,7 +436,8 @@ def _world8_pipeline_worker(rank, world, port, q):
 def test_world8_full_pipeline_drill():
     ctx = mp.get_context("spawn")
     world = 8
-    port = 29500 + ((os.getpid() + 13) % 400)
+    _PORT_SALT[0] += 1
+    port = 20000 + ((os.getpid() * 13 + _PORT_SALT[0] * 101) % 20000)
     q = ctx.Queue()
     procs = [ctx.Process(target=_world8_pipeline_worker,
                          args=(rank, world, port, q)) for rank in range(world)]
