@@ -30,42 +30,55 @@ __global__ void k_sdca_pass(const long long* __restrict__ indptr,
                             float* __restrict__ alpha,
                             float* __restrict__ v,
                             long long nrows, float scale /* 1/(lambda*n) */) {
-    // r2: 16-LANE GROUPS, 4 concurrent samples per wave.  The r1 wave-
-    // per-sample shape wasted ~45% of its lanes on RCV1-length rows
-    // (74 nnz vs 64 lanes: a full second round with 10 live lanes) and
-    // exposed every gather round trip; groups cover a 74-nnz row in 5
-    // ~93%-utilized rounds and quadruple the samples in flight per wave.
-    // Still hogwild: groups own disjoint sample subsets, only v is shared.
     const int lane = threadIdx.x & 63;
-    const int gl = lane & 15;
-    const long long grp =
-        ((long long)blockIdx.x * 4 + (threadIdx.x >> 6)) * 4 + (lane >> 4);
-    const long long ngrps = (long long)gridDim.x * 16;
-    if (grp >= nrows) return;
-    for (long long s = grp; s < nrows; s += ngrps) {
-        const long long i = perm ? (long long)perm[s] : s;
-        const long long e0 = indptr[i];
-        const long long e1 = indptr[i + 1];
-        const float nsq = norms_sq[i];
-        if (e1 <= e0 || nsq == 0.0f) continue;
-        float part = 0.0f;
-        for (long long t = e0 + gl; t < e1; t += 16)
-            part += values[t] * v[indices[t]];
-#pragma unroll
-        for (int off = 8; off; off >>= 1)
-            part += __shfl_xor(part, off, 16);
-        const float yi = y[i];
-        const float grad = (1.0f - yi * part) / (nsq * scale);
-        const float a = alpha[i];
-        float a_new = a + grad;
-        a_new = a_new < 0.0f ? 0.0f : (a_new > 1.0f ? 1.0f : a_new);
-        const float dalpha = a_new - a;
-        if (dalpha != 0.0f) {
-            if (gl == 0) alpha[i] = a_new;
-            const float c = dalpha * yi * scale;
-            for (long long t = e0 + gl; t < e1; t += 16)
-                atomicAdd(&v[indices[t]], c * values[t]);
+    const long long wave = (long long)blockIdx.x * 4 + (threadIdx.x >> 6);
+    const long long nwaves = (long long)gridDim.x * 4;
+    if (wave >= nrows) return;
+
+    // The per-sample chain is 3-4 dependent memory round trips (perm ->
+    // indptr -> indices -> v); prefetching the NEXT sample's metadata and
+    // first 64 nonzeros while the current sample computes hides most of it
+    // (the v gather itself stays at compute time for hogwild freshness).
+    struct Meta {
+        long long i, e0, e1;
+        float nsq, yi, a, val;
+        int idx;
+    };
+    auto fetch = [&](long long s, Meta& m) {
+        if (s >= nrows) { m.e0 = m.e1 = 0; m.idx = -1; return; }
+        m.i = perm ? (long long)perm[s] : s;
+        m.e0 = indptr[m.i];
+        m.e1 = indptr[m.i + 1];
+        m.nsq = norms_sq[m.i];
+        m.yi = y[m.i];
+        m.a = alpha[m.i];
+        const long long t = m.e0 + lane;
+        m.idx = (t < m.e1) ? indices[t] : -1;
+        m.val = (t < m.e1) ? values[t] : 0.0f;
+    };
+
+    Meta cur, nxt;
+    fetch(wave, cur);
+    for (long long s = wave; s < nrows; s = s + nwaves) {
+        fetch(s + nwaves, nxt);   // in flight across the current compute
+        if (cur.e1 > cur.e0 && cur.nsq != 0.0f) {
+            float part = (cur.idx >= 0) ? cur.val * v[cur.idx] : 0.0f;
+            for (long long t = cur.e0 + 64 + lane; t < cur.e1; t += WAVE)
+                part += values[t] * v[indices[t]];
+            const float dot = wave_reduce_sum(part);
+            const float grad = (1.0f - cur.yi * dot) / (cur.nsq * scale);
+            float a_new = cur.a + grad;
+            a_new = a_new < 0.0f ? 0.0f : (a_new > 1.0f ? 1.0f : a_new);
+            const float dalpha = a_new - cur.a;
+            if (dalpha != 0.0f) {
+                if (lane == 0) alpha[cur.i] = a_new;
+                const float c = dalpha * cur.yi * scale;
+                if (cur.idx >= 0) atomicAdd(&v[cur.idx], c * cur.val);
+                for (long long t = cur.e0 + 64 + lane; t < cur.e1; t += WAVE)
+                    atomicAdd(&v[indices[t]], c * values[t]);
+            }
         }
+        cur = nxt;
     }
 }
 
@@ -94,13 +107,12 @@ extern "C" hipError_t fma_sdca_pass(
     const float* y, const float* norms_sq, const int* perm, float* alpha,
     float* v, long long nrows, float scale, hipStream_t stream) {
     if (nrows <= 0) return hipErrorInvalidValue;
-    // keep >=16 samples sequential per GROUP: full-width hogwild
-    // degenerates to a synchronous full-batch step and stalls dual
-    // convergence.  4 groups per wave, 4 waves per block.
-    long long groups = nrows / 16;
-    if (groups < 16) groups = 16;
-    if (groups > 65536) groups = 65536;
-    unsigned grid = (unsigned)((groups + 15) / 16);
+    // keep >=16 samples sequential per wave: full-width hogwild degenerates
+    // to a synchronous full-batch step and stalls dual convergence
+    long long waves = nrows / 16;
+    if (waves < 4) waves = 4;
+    if (waves > 8192) waves = 8192;
+    unsigned grid = (unsigned)((waves + 3) / 4);
     k_sdca_pass<<<dim3(grid), dim3(256), 0, stream>>>(
         indptr, indices, values, y, norms_sq, perm, alpha, v, nrows, scale);
     return hipGetLastError();
