@@ -1028,8 +1028,15 @@ __global__ void k_als_solve_wavefused(const long long* __restrict__ indptr,
     constexpr int NA = KT * (KT + 1) / 2;
     constexpr int TROW = FP8 ? Geo<KT>::TROW8 : Geo<KT>::TROW;
     constexpr int SB = (K + 16) * TROW;            // stage bytes per wave
+    // split-wave staging: a chunk's gather tasks only need NT lanes (fp8
+    // K=64: 32); when two chunks' tasks fit in one wave, the upper half
+    // stages chunk ch+1 into a second buffer in the SAME instructions —
+    // double the gathers in flight at zero extra registers or occupancy
+    constexpr int NT = FP8 ? 8 * (K / 16) : K;
+    constexpr int NCHW = (2 * NT <= WAVE && NT >= 8) ? 2 : 1;
     constexpr int SCRF = (K * 17 > 1040) ? K * 17 : 1040;  // solver scratch
-    constexpr int WB = (SB > SCRF * 4) ? SB : SCRF * 4;    // union, per wave
+    constexpr int WB0 = (NCHW * SB > SCRF * 4) ? NCHW * SB : SCRF * 4;
+    constexpr int WB = (WB0 + 15) & ~15;
     __shared__ __align__(16) char smem[4 * WB];
     const int w = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
     const int lane = threadIdx.x & 63;
@@ -1049,11 +1056,13 @@ __global__ void k_als_solve_wavefused(const long long* __restrict__ indptr,
         }
         return;
     }
-    {   // zero my buffer's EXT pad rows K+2..K+15 (wave-local)
+    {   // zero the EXT pad rows K+2..K+15 of my NCHW buffers (wave-local)
         constexpr int SEGS = FP8 ? 8 : 16;
-        for (int i = lane; i < 14 * SEGS; i += WAVE) {
-            const int row = K + 2 + i / SEGS, seg = i % SEGS;
-            *(unsigned*)(buf + (long long)row * TROW + seg * 4) = 0u;
+        for (int i = lane; i < NCHW * 14 * SEGS; i += WAVE) {
+            const int b = i / (14 * SEGS), r2 = i % (14 * SEGS);
+            const int row = K + 2 + r2 / SEGS, seg = r2 % SEGS;
+            *(unsigned*)(buf + b * SB + (long long)row * TROW + seg * 4) =
+                0u;
         }
     }
     f32x4 T[NA], E[KT];
@@ -1062,15 +1071,36 @@ __global__ void k_als_solve_wavefused(const long long* __restrict__ indptr,
 #pragma unroll
     for (int p = 0; p < KT; ++p) E[p] = f32x4{0, 0, 0, 0};
     const int nchunks = (n + 31) >> 5;
-    for (int ch = 0; ch < nchunks; ++ch) {
-        stage_chunk_w<KT, FP8>(buf, indices, values, factors,
-                               p0 + (long long)ch * 32, n - ch * 32, lane);
+    for (int ch = 0; ch < nchunks; ch += NCHW) {
+        if constexpr (NCHW == 2) {
+            const int sub = (lane >= NT) ? 1 : 0;
+            const int lch = ch + sub;
+            if (lch < nchunks && lane < 2 * NT)
+                stage_chunk_w<KT, FP8, NT>(buf + sub * SB, indices, values,
+                                           factors,
+                                           p0 + (long long)lch * 32,
+                                           n - lch * 32, lane - sub * NT);
+        } else {
+            stage_chunk_w<KT, FP8>(buf, indices, values, factors,
+                                   p0 + (long long)ch * 32, n - ch * 32,
+                                   lane);
+        }
         WREG_FENCE();                  // cross-lane LDS write -> read
-        typename FragT<FP8>::type frag[KT + 1];
-        read_frags<KT, FP8>(buf, lane, frag);
+        {
+            typename FragT<FP8>::type frag[KT + 1];
+            read_frags<KT, FP8>(buf, lane, frag);
+            mfma_lower_tiles<KT, FP8>(frag, T);
+            mfma_ext_tiles<KT, FP8>(frag, E);
+        }
+        if constexpr (NCHW == 2) {
+            if (ch + 1 < nchunks) {
+                typename FragT<FP8>::type frag[KT + 1];
+                read_frags<KT, FP8>(buf + SB, lane, frag);
+                mfma_lower_tiles<KT, FP8>(frag, T);
+                mfma_ext_tiles<KT, FP8>(frag, E);
+            }
+        }
         WREG_FENCE();                  // reads drained before next stage
-        mfma_lower_tiles<KT, FP8>(frag, T);
-        mfma_ext_tiles<KT, FP8>(frag, E);
     }
     // b = bhi + blo redistributed to lane = row; then lambda*n*I
     wavefused_dump_b<KT>(E, scr, g4, li);
