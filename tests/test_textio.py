@@ -120,3 +120,55 @@ def test_params_negative_number_values():
     assert p.get_int("seed") == -7
     assert p.get_bool("flag") is True
     assert p.get("name") == "x"
+
+
+# ---------------------------------------------------- property tests
+
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+
+@settings(max_examples=500, deadline=None)
+@given(st.floats(allow_nan=False, allow_infinity=False))
+def test_java_double_round_trips(x):
+    """Shortest-round-trip property: parsing the Java-formatted string
+    recovers the exact double (JLS Double.toString contract)."""
+    s = t.java_double_to_string(x)
+    assert float(s) == x
+    # surface-shape invariants
+    assert "e" not in s or "E" in s  # exponents are uppercase E
+    if 1e-3 <= abs(x) < 1e7 or x == 0.0:
+        assert "E" not in s          # plain decimal in the JLS window
+
+
+@settings(max_examples=200, deadline=None)
+@given(st.integers(min_value=0, max_value=2**40),
+       st.sampled_from(["U", "I"]),
+       st.lists(st.floats(allow_nan=False, allow_infinity=False,
+                          width=32), min_size=1, max_size=16))
+def test_als_row_round_trip_property(eid, kind, vec):
+    row = t.als_factor_row(eid, kind, vec)
+    pid, pkind, pvec = t.parse_als_row(row)
+    assert pid == str(eid) and pkind == kind
+    assert pvec == pytest.approx(vec, rel=0, abs=0)  # exact round trip
+
+
+@settings(max_examples=100, deadline=None)
+@given(st.lists(st.tuples(
+    st.integers(min_value=0, max_value=10**6),
+    st.sampled_from(["U", "I"]),
+    st.lists(st.floats(allow_nan=False, allow_infinity=False, width=32),
+             min_size=3, max_size=3)), min_size=1, max_size=20))
+def test_native_parser_matches_python_parser(rows):
+    """The threaded C++ block parser agrees with the scalar Python codec
+    on arbitrary well-formed rows (fp32-exact values)."""
+    from flink_ms_amd import _hip_ops
+    text = "\n".join(t.als_factor_row(e, k, v) for e, k, v in rows)
+    ids, kinds, facs, offs, lens, bad = _hip_ops.parse_als_block(
+        text.encode(), 3)
+    assert int(bad) == 0
+    for r, (eid, kind, vec) in enumerate(rows):
+        assert int(ids[r]) == eid
+        assert ("U" if int(kinds[r]) == 0 else "I") == kind
+        for a, b in zip(facs[r].tolist(), vec):
+            assert a == b, (a, b)
