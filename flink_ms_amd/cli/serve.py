@@ -93,6 +93,10 @@ def main(argv=None) -> int:
                 n = torch.cuda.device_count()
                 if n > 0:
                     pd["device"] = f"cuda:{s % n}"
+            # per-shard native KvState port (kvPort, kvPort+1, ...; 0 stays
+            # 0 = each shard picks an ephemeral port and logs it)
+            if params.has("kvPort") and params.get_int("kvPort", 0) != 0:
+                pd["kvPort"] = str(params.get_int("kvPort") + s)
             procs.append(ctx.Process(target=_run_shard,
                                      args=(pd, base + s), daemon=True))
         for p in procs:

@@ -865,3 +865,25 @@ def test_kvserver_survives_malformed_requests():
         assert json.loads(r.read())["value"][1] == "0.5;0.5"
     finally:
         kv.stop()
+
+
+@pytest.mark.gpu
+def test_ingest_bulk_file_spill_gpu(tmp_path):
+    """Spill-mode bulk load on the real device: the bf16 HBM mirror is the
+    only dense factor copy; batched predict + SGD still work."""
+    path = tmp_path / "m.txt"
+    path.write_text("\n".join(
+        [f"{i},U,0.5;0.25;1.5;0.125" for i in range(5000)]
+        + [f"{i},I,1.0;0.5;0.25;2.0" for i in range(2000)]) + "\n")
+    store = ALSModelStore(device=torch.device("cuda:0"))
+    n = store.ingest_bulk_file(str(path))
+    assert n == 7000
+    assert "U" not in store._blocks.host
+    assert store._blocks.dev["U"].is_cuda
+    preds, ok = store.predict_batch(["7", "4999"], ["3", "1999"])
+    exp = 0.5 * 1.0 + 0.25 * 0.5 + 1.5 * 0.25 + 0.125 * 2.0
+    assert ok.tolist() == [True, True]
+    assert preds[0].item() == pytest.approx(exp, rel=2e-2)
+    b, sc, _ = store.sgd_update_batch(["7"], ["3"], [5.0],
+                                      learning_rate=0.1)
+    assert b == 1 and store.get_vector("7-U") != [0.5, 0.25, 1.5, 0.125]
