@@ -430,3 +430,55 @@ def test_wavefused2_matches_block_fused(gpu, k):
     torch.cuda.synchronize()
     assert out3[0].abs().sum() == 0 and out3[2].abs().sum() == 0
     assert torch.isfinite(out3[1]).all()
+
+
+def test_wavefused_torture_parity(gpu):
+    """Randomized shape/skew sweep of the flagship kernel: power-law
+    degrees, empty rows, single-rating rows, hot columns, odd tails, all
+    k in {16,32,48,64} x {bf16, fp8} — each config checked against the
+    fp32 reference on identically quantized inputs."""
+    g = torch.Generator().manual_seed(123)
+    for trial in range(6):
+        k = [16, 32, 48, 64][trial % 4]
+        rows = int(torch.randint(33, 700, (1,), generator=g))
+        cols = int(torch.randint(17, 400, (1,), generator=g))
+        # power-law-ish degrees with empty rows mixed in
+        deg = (torch.rand(rows, generator=g) ** 3 * 96).long()
+        deg[torch.rand(rows, generator=g) < 0.1] = 0
+        r_idx = torch.repeat_interleave(torch.arange(rows), deg)
+        nnz = r_idx.numel()
+        if nnz == 0:
+            continue
+        # hot columns: 30% of nnz hit 3 columns
+        c_idx = torch.randint(0, cols, (nnz,), generator=g)
+        hot = torch.rand(nnz, generator=g) < 0.3
+        c_idx[hot] = torch.randint(0, min(3, cols), (int(hot.sum()),),
+                                   generator=g)
+        vals = torch.rand(nnz, generator=g) * 4.5 + 0.5
+        csr = csr_from_coo(r_idx.int(), c_idx.int(), vals, rows, cols
+                           ).to(gpu)
+        fac32 = torch.randn(cols, k, generator=g) * 0.5
+        for dt in ("bf16", "fp8"):
+            if dt == "fp8":
+                fac = ops.quantize_fp8(fac32).to(gpu)
+                ref_fac = ops.dequantize_fp8(ops.quantize_fp8(fac32))
+                ref_vals = ops.fp8_rating_pair(vals)
+            else:
+                fac = fac32.to(torch.bfloat16).to(gpu)
+                ref_fac = fac32.to(torch.bfloat16).to(torch.float32)
+                hi = vals.to(torch.bfloat16).to(torch.float32)
+                ref_vals = hi + (vals - hi).to(torch.bfloat16).to(
+                    torch.float32)
+            out = ops.als_solve_side(csr, fac, reg=0.3)
+            cpu_csr = csr.to("cpu")
+            ref_csr = CSR(cpu_csr.indptr, cpu_csr.indices, ref_vals,
+                          rows, cols)
+            ref = R.als_solve_side_reference(ref_csr, ref_fac, reg=0.3)
+            torch.cuda.synchronize()
+            scale = max(1.0, float(ref.abs().amax()))
+            err = (out.cpu() - ref).abs().amax() / scale
+            assert err < 5e-3, (trial, k, dt, rows, cols, nnz, float(err))
+            # empty rows exactly zero
+            z = (deg == 0).nonzero(as_tuple=True)[0]
+            if z.numel():
+                assert out.cpu()[z].abs().sum() == 0.0
