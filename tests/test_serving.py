@@ -887,3 +887,62 @@ def test_ingest_bulk_file_spill_gpu(tmp_path):
     b, sc, _ = store.sgd_update_batch(["7"], ["3"], [5.0],
                                       learning_rate=0.1)
     assert b == 1 and store.get_vector("7-U") != [0.5, 0.25, 1.5, 0.125]
+
+
+def test_full_production_path_integration(tmp_path):
+    """Everything at once: train -> attach -> serve with fs WAL + native
+    kv plane -> bulk hot-swap -> batched SGD -> crash -> restore; every
+    surface stays coherent."""
+    import json
+    import urllib.request
+
+    from flink_ms_amd import _hip_ops
+    from flink_ms_amd.serving.app import create_app
+    u, i, r = synthetic_ratings(RatingsShape(60, 30, 900), seed=12)
+    model, _ = train_als(u, i, r, 60, 30,
+                         ALSConfig(iterations=3, num_factors=8,
+                                   lambda_=0.1, dtype=torch.float32))
+    uri = str(tmp_path / "ck")
+    kv = _hip_ops.KvServer()
+    kv_port = kv.start(0)
+    store = ALSModelStore(device=torch.device("cpu"))
+    store.attach_factors(model.user_factors, model.item_factors,
+                         model.user_ids, model.item_ids)
+    app = create_app(store, SVMModelStore(), checkpoint_data_uri=uri,
+                     checkpoint_interval_ms=0, state_backend="fs",
+                     kv_server=kv)
+    try:
+        with TestClient(app) as c:
+            # attached model visible on BOTH planes (startup kv sync)
+            p_http = c.get("/als/predict",
+                           params={"user": "3", "item": "5"}).json()
+            p_kv = json.loads(urllib.request.urlopen(
+                f"http://127.0.0.1:{kv_port}/als/predict?user=3&item=5"
+            ).read())
+            assert p_http["found"] and p_kv["found"]
+            assert p_kv["prediction"] == pytest.approx(
+                p_http["prediction"], rel=1e-12)
+            # bulk hot-swap of one row + batched SGD, then checkpoint
+            c.post("/model/als/rows",
+                   json={"rows": ["3,U,1.0;0;0;0;0;0;0;0",
+                                  "MEAN,U,0.1;0.1;0.1;0.1;0.1;0.1;0.1;0.1",
+                                  "MEAN,I,0.1;0.1;0.1;0.1;0.1;0.1;0.1;0.1"]})
+            c.post("/sgd/update_batch",
+                   json={"ratings": ["3\t5\t4.0"], "learning_rate": 0.1})
+            v_live = c.get("/state/ALS_MODEL/3-U").json()["value"][1]
+            assert v_live != "1.0;0;0;0;0;0;0;0"
+            c.post("/checkpoint")
+            # post-snapshot update (only in the WAL)
+            c.post("/sgd/update", json={"ratings": ["7\t2\t5.0"]})
+            v7 = c.get("/state/ALS_MODEL/7-U").json()["value"][1]
+    finally:
+        kv.stop()
+    # crash + restore: snapshot + WAL replay reproduce the full state
+    app2 = create_app(ALSModelStore(device=torch.device("cpu")),
+                      SVMModelStore(), checkpoint_data_uri=uri,
+                      checkpoint_interval_ms=0, state_backend="fs")
+    with TestClient(app2) as c:
+        assert c.get("/state/ALS_MODEL/3-U").json()["value"][1] == v_live
+        assert c.get("/state/ALS_MODEL/7-U").json()["value"][1] == v7
+        # the whole attached model survived via the snapshot
+        assert c.get("/state/ALS_MODEL/59-U").status_code == 200
