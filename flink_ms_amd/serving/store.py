@@ -637,7 +637,17 @@ class ALSModelStore:
             return out
 
     def __len__(self) -> int:
-        return len(self._payload)
+        """Number of keyed-state entries (payload rows plus attach/bulk
+        block rows not shadowed by a payload) — the consumer's keyed-state
+        cardinality, NOT just the queried subset."""
+        with self._lock:
+            n = len(self._payload)
+            if self._blocks is not None:
+                for kind in ("U", "I"):
+                    for eid in self._blocks.idmap[kind]:
+                        if f"{eid}-{kind}" not in self._payload:
+                            n += 1
+            return n
 
 
 class SVMModelStore:
