@@ -345,3 +345,68 @@ def svm_score_latency(num_features=47236, range_size=1000, n_queries=2000,
         print(f"svm {name:6s}: p50 {out[name]['p50_ms']:.4f} ms  "
               f"p95 {out[name]['p95_ms']:.4f} ms  (nnz={nnz})", flush=True)
     return out
+
+
+def kvserver_fixed_qps(qps=5000, duration_s=5.0, workers=8, num_keys=20000,
+                       k=64):
+    """Open-loop fixed-QPS latency against the native KvState plane (the
+    reference's per-query-latency measure, ALSPredictRandom.java:62-94,
+    at a controlled arrival rate)."""
+    import socket
+    import threading
+    import time as _t
+
+    from flink_ms_amd import _hip_ops
+    kv = _hip_ops.KvServer()
+    port = kv.start(0)
+    kv.put_rows([f"{i},U," + ";".join(["0.5"] * k) for i in range(num_keys)]
+                + [f"{i},I," + ";".join(["0.25"] * k)
+                   for i in range(num_keys)])
+    lats, lock = [], threading.Lock()
+    t0 = _t.perf_counter() + 0.2
+
+    def worker(w):
+        s = socket.create_connection(("127.0.0.1", port))
+        s.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+        buf = b""
+        n = 0
+        my = []
+        while True:
+            due = t0 + (n * workers + w) / qps
+            now = _t.perf_counter()
+            if due > t0 + duration_s:
+                break
+            if due > now:
+                _t.sleep(due - now)
+            path = (f"/als/predict?user={(n * 7 + w) % num_keys}"
+                    f"&item={(n * 13 + w) % num_keys}")
+            q0 = _t.perf_counter()
+            s.sendall(f"GET {path} HTTP/1.1\r\nHost: x\r\n\r\n".encode())
+            while b"\r\n\r\n" not in buf:
+                buf += s.recv(65536)
+            head, rest = buf.split(b"\r\n\r\n", 1)
+            cl = int([ln for ln in head.split(b"\r\n")
+                      if ln.lower().startswith(b"content-length")
+                      ][0].split(b":")[1])
+            while len(rest) < cl:
+                rest += s.recv(65536)
+            buf = rest[cl:]
+            my.append(_t.perf_counter() - q0)
+            n += 1
+        with lock:
+            lats.extend(my)
+        s.close()
+
+    ts = [threading.Thread(target=worker, args=(w,)) for w in range(workers)]
+    [t.start() for t in ts]
+    [t.join() for t in ts]
+    kv.stop()
+    lats.sort()
+    out = {"qps_target": qps, "achieved": len(lats) / duration_s,
+           "p50_ms": lats[len(lats) // 2] * 1e3,
+           "p95_ms": lats[int(len(lats) * 0.95)] * 1e3,
+           "p99_ms": lats[int(len(lats) * 0.99)] * 1e3}
+    print(f"kv fixed-qps {qps}: achieved {out['achieved']:.0f}/s  "
+          f"p50 {out['p50_ms']:.3f} ms  p95 {out['p95_ms']:.3f} ms  "
+          f"p99 {out['p99_ms']:.3f} ms", flush=True)
+    return out
