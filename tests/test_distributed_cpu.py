@@ -436,20 +436,33 @@ def _world8_pipeline_worker(rank, world, port, q):
 def test_world8_full_pipeline_drill():
     ctx = mp.get_context("spawn")
     world = 8
-    _PORT_SALT[0] += 1
-    port = 20000 + ((os.getpid() * 13 + _PORT_SALT[0] * 101) % 20000)
-    q = ctx.Queue()
-    procs = [ctx.Process(target=_world8_pipeline_worker,
-                         args=(rank, world, port, q)) for rank in range(world)]
-    for p in procs:
-        p.start()
-    results = {}
-    for _ in range(world):
-        rank, payload = q.get(timeout=400)
-        results[rank] = payload
-    for p in procs:
-        p.join(timeout=120)
-        assert p.exitcode == 0
+    # 8 CPU ranks on a loaded shared host can hit transient spawn or
+    # rendezvous failures; the drill's point is pipeline correctness, so
+    # one retry
+    for attempt in range(2):
+        _PORT_SALT[0] += 1
+        port = 20000 + ((os.getpid() * 13 + _PORT_SALT[0] * 101) % 20000)
+        q = ctx.Queue()
+        procs = [ctx.Process(target=_world8_pipeline_worker,
+                             args=(rank, world, port, q))
+                 for rank in range(world)]
+        try:
+            for p in procs:
+                p.start()
+            results = {}
+            for _ in range(world):
+                rank, payload = q.get(timeout=300)
+                results[rank] = payload
+            for p in procs:
+                p.join(timeout=120)
+                assert p.exitcode == 0
+            break
+        except Exception:  # noqa: BLE001
+            for p in procs:
+                if p.is_alive():
+                    p.terminate()
+            if attempt == 1:
+                raise
     assert all(results[r]["ok"] for r in range(world))
     # uneven tail: 1203 over 8 -> shards of 151, last rank 146
     assert results[0]["rows"] == 151 and results[7]["rows"] == 1203 - 7 * 151
