@@ -106,3 +106,74 @@ class ShardedQueryClient:
 
     def __exit__(self, *a):
         self.close()
+
+
+class ShardedKvClient:
+    """Key-routed point lookups over N native KvState servers (raw
+    keep-alive sockets; the Netty-KvStateServer-per-TaskManager topology
+    with `shard_of` as the location lookup).  ALS predict does its two
+    routed state fetches and the dot client-side, exactly like the
+    reference client (ALSPredict.java:69-83) — the shards may hold
+    disjoint key ranges, so the server-side /als/predict shortcut only
+    applies at one shard."""
+
+    def __init__(self, ports: List[int], host: str = "127.0.0.1",
+                 timeout_s: float = 5.0):
+        import socket
+        self._socks = []
+        self._bufs = []
+        for p in ports:
+            s = socket.create_connection((host, p), timeout=timeout_s)
+            s.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+            self._socks.append(s)
+            self._bufs.append(b"")
+        self.n = len(ports)
+
+    def _get(self, shard: int, path: str):
+        import json
+        s = self._socks[shard]
+        s.sendall(f"GET {path} HTTP/1.1\r\nHost: x\r\n\r\n".encode())
+        buf = self._bufs[shard]
+        while b"\r\n\r\n" not in buf:
+            buf += s.recv(65536)
+        head, rest = buf.split(b"\r\n\r\n", 1)
+        cl = int([ln for ln in head.split(b"\r\n")
+                  if ln.lower().startswith(b"content-length")][0]
+                 .split(b":")[1])
+        while len(rest) < cl:
+            rest += s.recv(65536)
+        self._bufs[shard] = rest[cl:]
+        status = int(head.split(b" ", 2)[1])
+        return status, json.loads(rest[:cl])
+
+    def query_state(self, name: str, key: str) -> Optional[Tuple[str, str]]:
+        st, body = self._get(shard_of(key, self.n),
+                             f"/state/{name}/{key}")
+        if st != 200:
+            return None
+        return tuple(body["value"])
+
+    def als_predict(self, user: str, item: str) -> dict:
+        u = self.query_state("ALS_MODEL", f"{user}-U")
+        v = self.query_state("ALS_MODEL", f"{item}-I")
+        if u is None or v is None:
+            return {"found": False,
+                    "message": "User or Item Factors do not exist in the "
+                               f"model for the query: {user},{item}"}
+        uv = [float(x) for x in u[1].split(";")]
+        vv = [float(x) for x in v[1].split(";")]
+        return {"found": True,
+                "prediction": sum(a * b for a, b in zip(uv, vv))}
+
+    def close(self):
+        for s in self._socks:
+            try:
+                s.close()
+            except OSError:
+                pass
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        self.close()

@@ -946,3 +946,38 @@ def test_full_production_path_integration(tmp_path):
         assert c.get("/state/ALS_MODEL/7-U").json()["value"][1] == v7
         # the whole attached model survived via the snapshot
         assert c.get("/state/ALS_MODEL/59-U").status_code == 200
+
+
+def test_sharded_kv_plane_routing():
+    """N native KvState servers + crc32 key routing: rows land on their
+    shard only, routed lookups and client-side predicts agree with an
+    unsharded store."""
+    from flink_ms_amd import _hip_ops
+    from flink_ms_amd.serving.sharding import (ShardedKvClient, als_row_key,
+                                               shard_of)
+    n = 3
+    servers = [_hip_ops.KvServer() for _ in range(n)]
+    ports = [kv.start(0) for kv in servers]
+    rows = [f"{i},U,0.5;{i}.25" for i in range(40)]
+    rows += [f"{i},I,1.5;0.5" for i in range(40)]
+    ref = ALSModelStore(device=torch.device("cpu"))
+    ref.ingest(rows)
+    for row in rows:   # producer-side routing
+        servers[shard_of(als_row_key(row), n)].put_rows([row])
+    try:
+        assert sum(kv.size() for kv in servers) == len(rows)
+        assert max(kv.size() for kv in servers) < len(rows)  # actually split
+        with ShardedKvClient(ports) as sc:
+            for key in ("0-U", "17-U", "39-I", "MEANX-U"):
+                got = sc.query_state("ALS_MODEL", key)
+                exp = ref.query(key)
+                assert got == exp, key
+            for u, i in (("3", "7"), ("39", "0")):
+                p = sc.als_predict(u, i)
+                assert p["found"]
+                assert p["prediction"] == pytest.approx(
+                    ref.predict(u, i), rel=1e-12)
+            assert not sc.als_predict("404", "0")["found"]
+    finally:
+        for kv in servers:
+            kv.stop()
