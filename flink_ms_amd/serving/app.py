@@ -251,6 +251,10 @@ def create_app(als_store: Optional[ALSModelStore] = None,
         # parser + ONE H2D mirror slab (vs row-at-a-time /model/als/rows).
         # spill=true (single file): mmap-backed larger-than-memory load —
         # factors live in the device mirror + byte slices of the file.
+        # NOTE durability: spill loads are NOT WAL-journaled (the source
+        # file IS the durable artifact; journaling a 15 GB model would
+        # defeat the point) — after a crash, re-issue the spill load, then
+        # the WAL replays the post-load deltas on top.
         if body.spill and os.path.isfile(body.path):
             try:
                 n = als.ingest_bulk_file(body.path)
