@@ -23,6 +23,8 @@
 #include <unistd.h>
 
 #include <atomic>
+#include <cerrno>
+#include <chrono>
 #include <cstring>
 #include <memory>
 #include <shared_mutex>
@@ -89,6 +91,11 @@ public:
         shutdown(listen_fd_, SHUT_RDWR);
         close(listen_fd_);
         if (acceptor_.joinable()) acceptor_.join();
+        // connection threads poll running_ via a 200 ms recv timeout;
+        // wait (bounded) until they all exit so the store cannot be
+        // destroyed under a live reader
+        for (int i = 0; i < 25 && conns_.load() > 0; ++i)
+            std::this_thread::sleep_for(std::chrono::milliseconds(100));
     }
 
     // rows: "<id>,<U|I>,<f;f;...>" — parsed once here, then both the
@@ -141,7 +148,13 @@ private:
             }
             int one = 1;
             setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
-            std::thread([this, fd] { serve_conn(fd); }).detach();
+            timeval tv{0, 200000};   // recv timeout: running_ poll period
+            setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
+            conns_.fetch_add(1);
+            std::thread([this, fd] {
+                serve_conn(fd);
+                conns_.fetch_sub(1);
+            }).detach();
         }
     }
 
@@ -242,10 +255,18 @@ private:
             size_t hdr_end;
             while ((hdr_end = buf.find("\r\n\r\n")) == std::string::npos) {
                 ssize_t r = recv(fd, tmp, sizeof(tmp), 0);
-                if (r <= 0) { close(fd); return; }
+                if (r == 0) { close(fd); return; }
+                if (r < 0) {
+                    if ((errno == EAGAIN || errno == EWOULDBLOCK ||
+                         errno == EINTR) && running_)
+                        continue;   // timeout tick: re-check running_
+                    close(fd);
+                    return;
+                }
                 buf.append(tmp, (size_t)r);
                 if (buf.size() > 1 << 20) { close(fd); return; }
             }
+            if (!running_) { close(fd); return; }
             // request line: METHOD SP PATH SP VERSION
             size_t sp1 = buf.find(' ');
             size_t sp2 = buf.find(' ', sp1 + 1);
@@ -265,6 +286,7 @@ private:
     int listen_fd_ = -1;
     int port_ = 0;
     std::atomic<bool> running_{false};
+    std::atomic<int> conns_{0};
     std::thread acceptor_;
 };
 
