@@ -401,3 +401,55 @@ def test_temporary_path_stages_iterations(tmp_path):
     # the staged final iteration matches the written model
     assert open(stage / "iteration-2" / "userFactors").read() == \
         open(tmp_path / "uf").read()
+
+
+@pytest.mark.timeout(300)
+def test_serve_cli_kv_plane_subprocess(tmp_path):
+    """`serve --kvPort 0`: the native KvState server comes up next to the
+    FastAPI app, gets the preloaded model pushed, and answers the hot GET
+    surface compatibly."""
+    import json
+    import re
+    import subprocess
+    import sys
+    import urllib.request
+
+    model = tmp_path / "m.model"
+    model.write_text("1,U,0.5;0.25\n2,I,0.75;0.125\n")
+    port = _free_port()
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "flink_ms_amd.cli.serve",
+         "--port", str(port), "--device", "cpu", "--kvPort", "0",
+         "--alsModel", str(model)],
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
+    try:
+        kv_port = None
+        deadline = time.time() + 60
+        while time.time() < deadline and kv_port is None:
+            line = proc.stdout.readline()
+            m = re.search(r"kvserver\] native KvState server on "
+                          r"127\.0\.0\.1:(\d+)", line)
+            if m:
+                kv_port = int(m.group(1))
+        assert kv_port, "kvserver did not announce a port"
+        # wait for readiness, then query the data plane directly
+        for _ in range(200):
+            try:
+                r = urllib.request.urlopen(
+                    f"http://127.0.0.1:{kv_port}/healthz", timeout=2)
+                break
+            except Exception:
+                time.sleep(0.1)
+        body = json.loads(urllib.request.urlopen(
+            f"http://127.0.0.1:{kv_port}/state/ALS_MODEL/1-U",
+            timeout=5).read())
+        assert body["value"] == ["1-U", "0.5;0.25"]
+        pred = json.loads(urllib.request.urlopen(
+            f"http://127.0.0.1:{kv_port}/als/predict?user=1&item=2",
+            timeout=5).read())
+        assert pred["found"]
+        assert pred["prediction"] == pytest.approx(
+            0.5 * 0.75 + 0.25 * 0.125, rel=1e-12)
+    finally:
+        proc.terminate()
+        proc.wait(timeout=30)
