@@ -367,6 +367,12 @@ def test_chunked_allgather_remap_unit():
             replica.extend(shard_of[r][a:b])
     for gid in range(22):
         assert replica[int(pos[gid])] == gid, gid
+    # degenerate shapes: more chunks than shard rows, and a single chunk
+    for nch in (1, 50):
+        ch2 = ChunkedAllgather(ctx, part, chunks=nch)
+        pos2 = ch2.remap_indices(ids)
+        assert len(set(pos2.tolist())) == 22
+        assert int(pos2.max()) < ch2.replica_rows
 
 
 def _als_overlap_worker(rank, world, port, q, chunks):
