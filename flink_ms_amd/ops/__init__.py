@@ -103,12 +103,18 @@ def als_solve_side(
     opposite side's factors.  Returns fp32 [num_rows, k]; optionally also
     writes the bf16 (or e4m3 ``out_fp8``) image for the next half-iteration.
 
-    GPU paths: the default modular path (MFMA Gramian kernel -> batched
-    wave-per-entity LDL solve; best measured occupancy at k <= 64) and the
-    single-launch fused kernel (``fused=True``/k > 64; avoids the A
-    round-trip through HBM but serializes the solve at Gramian occupancy).
+    GPU paths (fastest first; see docs/KERNELS.md for the measured ladder):
+    - k <= 64 DEFAULT: the wave-fused kernel — Gramian MFMAs accumulate
+      straight into the register LDL's lower-triangle C-fragments (one
+      wave per entity, no A in HBM, no barriers; split-wave dual-chunk
+      staging).  ``FMA_WAVEFUSED_DB=1`` selects the software-pipelined
+      ablation.
+    - k > 64 (or ``fused=True``): the block-fused kernel (triangular LDS
+      A image, 4 blocks/CU).
+    - ``slab_rows``: the modular gramian -> batched-solve path, kept for
+      parity tests and A-materializing consumers.
     ``other_factors`` dtype selects the gather precision: uint8 = e4m3
-    bytes (one cache line per k<=64 row — the r2 headline path), anything
+    bytes (one cache line per k<=64 row — the flagship path), anything
     else is cast to bf16.
     """
     if other_factors.is_cuda:
