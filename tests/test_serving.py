@@ -537,6 +537,7 @@ def test_bulk_ingest_throughput_smoke():
     scalar path ON THE SAME MACHINE (absolute rates vary wildly with
     shared-host load; the real 1.06M rows/s number is measured on the GPU
     box, benchmarks/bench_bulk_ingest.py)."""
+    pytest.importorskip("flink_ms_amd._hip_ops")
     import time as _t
     k = 16
     n = 50_000
@@ -693,7 +694,7 @@ def test_kvserver_parity_with_fastapi():
 
     from fastapi.testclient import TestClient
 
-    from flink_ms_amd import _hip_ops
+    _hip_ops = pytest.importorskip("flink_ms_amd._hip_ops")
     from flink_ms_amd.serving.app import create_app
     kv = _hip_ops.KvServer()
     port = kv.start(0)
@@ -755,6 +756,7 @@ def test_ingest_bulk_file_spill_mode(tmp_path):
     only in the device mirror + on-disk byte slices (no dense host copy),
     while queries, predicts, batched kernels, SGD write-backs and
     snapshots all stay correct."""
+    pytest.importorskip("flink_ms_amd._hip_ops")
     rows = [f"{i},U,0.5;0.25;{i % 7}.5" for i in range(200)]
     rows += [f"{i},I,1.0;0.125;0.75" for i in range(100)]
     path = tmp_path / "model.txt"
@@ -790,6 +792,7 @@ def test_ingest_bulk_file_spill_mode(tmp_path):
 def test_spill_then_attach_keeps_correct_vectors(tmp_path):
     """Once a kind spills (mmap bulk load), later keep-host blocks must
     NOT resurrect a zero-filled host tensor for the disk-backed rows."""
+    pytest.importorskip("flink_ms_amd._hip_ops")
     path = tmp_path / "m.txt"
     path.write_text("\n".join(f"{i},U,1.5;2.5" for i in range(50)) + "\n")
     store = ALSModelStore(device=torch.device("cpu"))
@@ -808,7 +811,7 @@ def test_parse_block_fuzz_matches_scalar_parser():
     parser on arbitrary well-formed rows, and flags malformed ones."""
     import random
 
-    from flink_ms_amd import _hip_ops
+    _hip_ops = pytest.importorskip("flink_ms_amd._hip_ops")
     from flink_ms_amd.utils.textio import parse_als_row
     rng = random.Random(11)
     k = 5
@@ -843,7 +846,7 @@ def test_kvserver_survives_malformed_requests():
     (the reference's Netty server has the same contract)."""
     import socket
 
-    from flink_ms_amd import _hip_ops
+    _hip_ops = pytest.importorskip("flink_ms_amd._hip_ops")
     kv = _hip_ops.KvServer()
     port = kv.start(0)
     kv.put_rows(["1,U,0.5;0.5"])
@@ -901,7 +904,7 @@ def test_full_production_path_integration(tmp_path):
     import json
     import urllib.request
 
-    from flink_ms_amd import _hip_ops
+    _hip_ops = pytest.importorskip("flink_ms_amd._hip_ops")
     from flink_ms_amd.serving.app import create_app
     u, i, r = synthetic_ratings(RatingsShape(60, 30, 900), seed=12)
     model, _ = train_als(u, i, r, 60, 30,
@@ -957,7 +960,7 @@ def test_sharded_kv_plane_routing():
     """N native KvState servers + crc32 key routing: rows land on their
     shard only, routed lookups and client-side predicts agree with an
     unsharded store."""
-    from flink_ms_amd import _hip_ops
+    _hip_ops = pytest.importorskip("flink_ms_amd._hip_ops")
     from flink_ms_amd.serving.sharding import (ShardedKvClient, als_row_key,
                                                shard_of)
     n = 3
