@@ -272,9 +272,12 @@ def test_driver_bench_launch_contract(tmp_path, world):
     import subprocess
     import sys
 
-    # world-8 CPU launches occasionally hit transient SIGABRTs on loaded
-    # shared hosts (gloo/TCPStore timing, not product logic) — one retry
-    for attempt in range(2):
+    # world-8 CPU launches occasionally hit transient worker deaths on
+    # loaded shared hosts (gloo/TCPStore timing, not product logic):
+    # pin gloo to loopback and retry
+    env = {**os.environ, "GLOO_SOCKET_IFNAME": "lo",
+           "TP_SOCKET_IFNAME": "lo"}
+    for attempt in range(3):
         res = subprocess.run(
             [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
              "--nproc-per-node", str(world), "--master-addr", "127.0.0.1",
@@ -284,7 +287,7 @@ def test_driver_bench_launch_contract(tmp_path, world):
              "--users-per-gpu", "300", "--items", "200",
              "--ratings-per-gpu", "5000", "--rank", "16",
              "--svm-rows-per-gpu", "400"],
-            capture_output=True, text=True, timeout=500)
+            capture_output=True, text=True, timeout=500, env=env)
         if res.returncode == 0:
             break
     assert res.returncode == 0, res.stderr[-2000:]
