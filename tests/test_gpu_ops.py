@@ -482,3 +482,41 @@ def test_wavefused_torture_parity(gpu):
             z = (deg == 0).nonzero(as_tuple=True)[0]
             if z.numel():
                 assert out.cpu()[z].abs().sum() == 0.0
+
+
+def test_wavefused_deterministic(gpu):
+    """The flagship kernel is atomics-free: identical inputs must produce
+    BITWISE identical outputs across launches (reproducible training)."""
+    csr = _rand_csr(rows=500, cols=300, nnz=40_000, seed=5, device=gpu)
+    fac = ops.quantize_fp8(
+        torch.randn(300, 64, generator=torch.Generator().manual_seed(1))
+        * 0.5).to(gpu)
+    a = ops.als_solve_side(csr, fac, reg=0.5)
+    b = ops.als_solve_side(csr, fac, reg=0.5)
+    torch.cuda.synchronize()
+    assert torch.equal(a, b)
+
+
+def test_als_gpu_default_rank10_fp8(gpu):
+    """The reference's DEFAULT numFactors is 10 (ALSImpl.scala) — a
+    non-multiple-of-16 rank must pad cleanly through the fp8 wave-fused
+    path and train."""
+    from flink_ms_amd.data.ratings import RatingsShape, synthetic_ratings
+    from flink_ms_amd.models.als import ALSConfig, ALSTrainer
+    from flink_ms_amd.models.mse import evaluate_mse
+    g = torch.Generator().manual_seed(31)
+    U0 = torch.randn(2000, 6, generator=g) * 0.5
+    V0 = torch.randn(800, 6, generator=g) * 0.5
+    u = torch.randint(0, 2000, (60_000,), generator=g)
+    i = torch.randint(0, 800, (60_000,), generator=g)
+    r = (U0[u] * V0[i]).sum(dim=1)
+    tr = ALSTrainer(ALSConfig(iterations=5, num_factors=10, lambda_=0.05,
+                              factor_dtype="fp8"))
+    tr.ctx.device = gpu
+    tr.setup(u.long(), i.long(), r, 2000, 800)
+    tr.fit()
+    m = tr.model()
+    assert m.user_factors.shape[1] == 10
+    res = evaluate_mse(m.user_factors.to(gpu), m.item_factors.to(gpu),
+                       u, i, r)
+    assert res.mse < 0.15 * float(r.var()), res.mse
