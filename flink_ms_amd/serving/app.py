@@ -319,8 +319,12 @@ def create_app(als_store: Optional[ALSModelStore] = None,
     def svm_predict(body: SVMPredictBody):
         pairs = []
         for tok in body.vector.strip().split():
-            fid, val = tok.split(":")
-            pairs.append((fid, float(val)))
+            try:
+                fid, val = tok.split(":")
+                pairs.append((fid, float(val)))
+            except ValueError:
+                raise HTTPException(
+                    400, f"malformed sparse-vector token: {tok!r}")
         pred, raw, messages = svm.predict(
             pairs, body.output_decision_function, body.threshold_value,
             body.range)

@@ -989,3 +989,13 @@ def test_sharded_kv_plane_routing():
     finally:
         for kv in servers:
             kv.stop()
+
+
+def test_svm_predict_malformed_vector_is_400(svm_store):
+    from flink_ms_amd.serving.app import create_app
+    app = create_app(ALSModelStore(device=torch.device("cpu")), svm_store)
+    with TestClient(app) as c:
+        r = c.post("/svm/predict", json={"vector": "1:0.5 oops 2:1.0"})
+        assert r.status_code == 400 and "oops" in r.json()["detail"]
+        r = c.post("/svm/predict", json={"vector": "1:notanum"})
+        assert r.status_code == 400
