@@ -37,14 +37,27 @@ def run_once(client: QueryClientHelper, rows, params, out_file=None) -> int:
     )
     if not rows:
         return 0
+    # --gpuBatch: route through the K4 batched device kernel
+    # (/sgd/update_batch) instead of the scalar fp64 loop — v1 semantics
+    # at the store's bf16 precision, MEAN cold-start preserved
+    if params.get_bool("gpuBatch", False):
+        def _send(rs):
+            return client.sgd_update_batch(
+                rs, field_delimiter=kw["field_delimiter"],
+                learning_rate=kw["learning_rate"],
+                user_regularization=kw["user_regularization"],
+                item_regularization=kw["item_regularization"])
+    else:
+        def _send(rs):
+            return client.sgd_update(rs, **kw)
     if params.get("outputMode", "kafka") == "hdfs":
         # compute on the server but persist rows to the output path
-        resp = client.sgd_update(rows, **kw)
+        resp = _send(rows)
         with open(params.get_required("outputPath"), "a") as f:
             for row in resp["rows"]:
                 f.write(row + "\n")
     else:
-        resp = client.sgd_update(rows, **kw)
+        resp = _send(rows)
     for rid in resp.get("nan_records", []):
         print(f"NaN detected for: {rid}")
     return resp["updated"]

@@ -72,6 +72,23 @@ class QueryClientHelper:
         r.raise_for_status()
         return r.json()
 
+    def sgd_update_batch(self, ratings, field_delimiter: str = "\t",
+                         learning_rate: float = 0.1,
+                         user_regularization: float = 0.0,
+                         item_regularization: float = 0.0) -> dict:
+        """Batched online SGD through the K4 device kernel
+        (/sgd/update_batch); unresolvable ids fall back to the scalar
+        MEAN-cold-start path server-side."""
+        r = self._client.post(f"{self.base}/sgd/update_batch", json={
+            "ratings": ratings,
+            "field_delimiter": field_delimiter,
+            "learning_rate": learning_rate,
+            "user_regularization": user_regularization,
+            "item_regularization": item_regularization,
+        })
+        r.raise_for_status()
+        return r.json()
+
     def close(self):
         self._client.close()
 

@@ -457,3 +457,23 @@ def test_serve_cli_kv_plane_subprocess(tmp_path):
     finally:
         proc.terminate()
         proc.wait(timeout=30)
+
+
+@pytest.mark.timeout(120)
+def test_sgd_cli_gpu_batch_mode(tmp_path, server):
+    """--gpuBatch routes the stream through /sgd/update_batch (K4 path)."""
+    port = server
+    with QueryClientHelper("127.0.0.1", port, 5) as c:
+        c.ingest_rows("als", ["910001,U,1.0;1.0", "920001,I,0.5;0.5",
+                              "MEAN,U,0.2;0.2", "MEAN,I,0.2;0.2"])
+    stream = tmp_path / "s.tsv"
+    stream.write_text("910001\t920001\t4.0\n999999\t920001\t2.0\n")
+    from flink_ms_amd.cli import sgd as sgd_cli
+    rc = sgd_cli.main(["--input", str(stream), "--mode", "once",
+                       "--jobId", "j", "--jobManagerPort", str(port),
+                       "--gpuBatch", "--learningRate", "0.1"])
+    assert rc == 0
+    with QueryClientHelper("127.0.0.1", port, 5) as c:
+        hit = c.query_state("ALS_MODEL", "910001-U")
+        assert hit is not None and hit[1] != "1.0;1.0"
+        assert c.query_state("ALS_MODEL", "999999-U") is not None  # cold
