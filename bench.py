@@ -137,7 +137,10 @@ def bench_als(args, ctx):
                                 and args.factor_dtype == "fp8" else
                                 ("bf16" if on_gpu else "fp32")),
             "solve_dtype": "fp32",
-            "parallelism": f"dp{world}+factor-allgather",
+            "parallelism": (f"dp{world}+factor-allgather"
+                            + ("+chunked-overlap"
+                               if getattr(trainer, "_overlap", False)
+                               else "")),
         },
     }
 

@@ -299,7 +299,8 @@ def test_driver_bench_launch_contract(tmp_path, world):
     assert d["scaling"] == "weak"
     assert d["config"]["global_batch"] == 5000 * world  # whole-job aggregate
     assert d["value"] > 0 and d["ms_per_step"] > 0
-    assert d["config"]["parallelism"] == f"dp{world}+factor-allgather"
+    assert d["config"]["parallelism"] == \
+        f"dp{world}+factor-allgather+chunked-overlap"
     assert d["config"]["svm_samples_per_sec"] > 0
 
 
