@@ -17,7 +17,7 @@ class HipccBuildExt(_build_ext):
 
 setup(
     name="flink_ms_amd",
-    version="0.1.0",
+    version="0.2.0",
     packages=["flink_ms_amd", "flink_ms_amd.cli", "flink_ms_amd.data",
               "flink_ms_amd.models", "flink_ms_amd.ops",
               "flink_ms_amd.parallel", "flink_ms_amd.serving",
