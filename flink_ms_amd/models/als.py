@@ -184,7 +184,17 @@ class ALSTrainer:
         self.user_f32: Optional[torch.Tensor] = None
         self.item_f32: Optional[torch.Tensor] = None
         if self._overlap:
-            self._prep_overlap(dev, kp)
+            try:
+                self._prep_overlap(dev, kp)
+            except Exception:
+                # graceful degradation: overlap SETUP failure (stream or
+                # replica allocation on an unforeseen runtime) falls back
+                # to the serial exchange — the run completes and the
+                # reported parallelism string reflects the actual mode.
+                # Step-time errors are NOT caught (they would mask bugs).
+                log.exception("overlapped-exchange setup failed; falling "
+                              "back to the serial exchange")
+                self._overlap = False
         log.info(
             "ALS setup: %d users x %d items, %d local ratings, k=%d, "
             "world=%d, exchange=%s",
